@@ -1,0 +1,107 @@
+"""Per-step HIP caching-allocator watermarks.
+
+On ROCm, ``torch.cuda.reset_peak_memory_stats`` / ``max_memory_allocated`` /
+``max_memory_reserved`` read the HIP caching allocator directly (the torch
+"cuda" namespace IS HIP on a ROCm build — there is no CUDA anywhere in the
+stack). Semantics mirrored from the reference (utils/step_memory.py:34-114):
+watermarks are reset at step start and read at step end; on a machine
+without a GPU the values are ``None`` (null means "not measured", never 0).
+
+The device-capacity field is carried alongside so pressure diagnosis scales
+to the MI355X's 288 GB HBM3E without hard-coded capacities anywhere else.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from collections import deque
+from dataclasses import dataclass
+from typing import Deque, List, Optional
+
+STEP_MEMORY_QUEUE_MAX = 4096
+
+
+@dataclass
+class StepMemoryEvent:
+    step: int
+    timestamp: float
+    peak_allocated_bytes: Optional[int]
+    peak_reserved_bytes: Optional[int]
+    device_capacity_bytes: Optional[int]
+    device: Optional[str]
+
+
+_queue_lock = threading.Lock()
+_queue: Deque[StepMemoryEvent] = deque()
+
+
+def _cuda():
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            return torch.cuda
+    except Exception:
+        pass
+    return None
+
+
+class StepMemoryTracker:
+    """Reset-at-start / record-at-end peak watermark tracker for one model."""
+
+    def __init__(self, model=None) -> None:
+        self._cuda = _cuda()
+        self._device_index: Optional[int] = None
+        self._capacity: Optional[int] = None
+        if self._cuda is not None:
+            try:
+                self._device_index = self._cuda.current_device()
+                props = self._cuda.get_device_properties(self._device_index)
+                self._capacity = int(props.total_memory)
+            except Exception:
+                self._cuda = None
+
+    def reset(self) -> None:
+        if self._cuda is None:
+            return
+        try:
+            self._cuda.reset_peak_memory_stats(self._device_index)
+        except Exception:
+            pass
+
+    def record(self, step: int) -> None:
+        peak_alloc: Optional[int] = None
+        peak_reserved: Optional[int] = None
+        device: Optional[str] = None
+        if self._cuda is not None:
+            try:
+                peak_alloc = int(self._cuda.max_memory_allocated(self._device_index))
+                peak_reserved = int(self._cuda.max_memory_reserved(self._device_index))
+                device = f"cuda:{self._device_index}"
+            except Exception:
+                peak_alloc = peak_reserved = None
+        event = StepMemoryEvent(
+            step=step,
+            timestamp=time.time(),
+            peak_allocated_bytes=peak_alloc,
+            peak_reserved_bytes=peak_reserved,
+            device_capacity_bytes=self._capacity,
+            device=device,
+        )
+        with _queue_lock:
+            if len(_queue) >= STEP_MEMORY_QUEUE_MAX:
+                _queue.popleft()
+            _queue.append(event)
+
+
+def drain_step_memory_queue() -> List[StepMemoryEvent]:
+    with _queue_lock:
+        out = list(_queue)
+        _queue.clear()
+    return out
+
+
+def clear_for_tests() -> None:
+    with _queue_lock:
+        _queue.clear()

@@ -1,0 +1,117 @@
+"""``trace_step``: the per-step bracket (reference: sdk/instrumentation.py:127-260).
+
+Enter: publish runtime environment once, reset HIP-allocator watermarks,
+open the ``step_time`` envelope event (CPU wall + device ring stamp), arm
+the per-thread phase enables + forward-target identity set, ensure
+optimizer hooks (auto mode).
+
+Exit (finally): close the envelope, disarm phase flags, advance the step
+counter (gradient-accumulation micro-steps count as steps), record memory
+watermarks, flush the step's events as one batch, notify the recording
+budget, and kick the RCCL rank-stats exchange when active.
+"""
+
+from __future__ import annotations
+
+import functools
+import time
+from contextlib import contextmanager
+
+from traceml_amd.core import event_names
+from traceml_amd.core.arming import is_tracing_armed, phase_flags
+from traceml_amd.core.flush import flush_step_events
+from traceml_amd.core.step_memory import StepMemoryTracker
+from traceml_amd.core.timing import TimeEvent, close_event, open_event, record_event
+from traceml_amd.instrumentation.hooks.optimizer_hooks import (
+    ensure_optimizer_timing_installed,
+)
+from traceml_amd.instrumentation.patches.forward import forward_target_ids
+from traceml_amd.runtime import environment, state
+from traceml_amd.runtime.identity import resolve_runtime_identity
+from traceml_amd.sdk import initial
+
+
+@contextmanager
+def trace_step(model=None):
+    config = initial.get_active_config()
+    recording = state.recording_state()
+    if (
+        config is None
+        or config.noop
+        or not is_tracing_armed()
+        or not recording.should_record_trace_events()
+    ):
+        yield
+        if config is not None and not config.noop:
+            state.session_state().advance()
+        return
+
+    identity = resolve_runtime_identity()
+    environment.publish_runtime_environment_once(identity, model)
+
+    mem_tracker = StepMemoryTracker(model)
+    mem_tracker.reset()
+
+    flags = phase_flags()
+    flags.in_step = True
+    flags.forward_targets = forward_target_ids(model) if model is not None else ()
+    flags.forward_enabled = config.patch_forward and model is not None
+    flags.backward_enabled = config.patch_backward
+    flags.h2d_enabled = config.patch_h2d
+    flags.optimizer_enabled = config.mode == "auto"
+    if config.auto_optimizer_hooks:
+        ensure_optimizer_timing_installed()
+
+    step_event = open_event(event_names.STEP_TIME)
+    try:
+        yield
+    finally:
+        close_event(step_event)
+        flags.in_step = False
+        flags.forward_enabled = False
+        flags.backward_enabled = False
+        flags.h2d_enabled = False
+        flags.optimizer_enabled = False
+        flags.forward_targets = ()
+        step = state.session_state().advance()
+        mem_tracker.record(step)
+        flush_step_events(step)
+        recording.mark_trace_step_flushed()
+        _kick_rank_stats(step)
+
+
+def _kick_rank_stats(step: int) -> None:
+    try:
+        from traceml_amd.parallel.rank_stats import get_active_exchange
+
+        exchange = get_active_exchange()
+        if exchange is not None:
+            exchange.on_step_flushed(step)
+    except Exception:
+        pass
+
+
+def trace_time(name: str):
+    """Decorator: time a function as a custom region inside the step."""
+
+    def decorator(fn):
+        @functools.wraps(fn)
+        def wrapper(*args, **kwargs):
+            if not is_tracing_armed():
+                return fn(*args, **kwargs)
+            cpu_start = time.time()
+            try:
+                return fn(*args, **kwargs)
+            finally:
+                record_event(
+                    TimeEvent(
+                        name=f"_traceml_user:{name}",
+                        device="cpu",
+                        cpu_start=cpu_start,
+                        cpu_end=time.time(),
+                    )
+                )
+
+        return wrapper
+
+    return decorator
