@@ -1,0 +1,172 @@
+"""Final report generator → ``final_summary.{json,txt}``
+(reference: reporting/final.py:47-990).
+
+Top-level shape (schema 1.7-compatible): schema_version / generated_at /
+duration_s / meta / primary_diagnosis / system / process / step_time /
+step_memory / text. ``text`` is the verdict-first compact report:
+TraceML Verdict, Why, Next, then the section cards.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+import sqlite3
+from typing import Optional
+
+from traceml_amd.reporting.primary import build_primary_diagnosis
+from traceml_amd.reporting.schema import SCHEMA_VERSION
+from traceml_amd.reporting.sections import process as process_section
+from traceml_amd.reporting.sections import step_memory as step_memory_section
+from traceml_amd.reporting.sections import step_time as step_time_section
+from traceml_amd.reporting.sections import system as system_section
+from traceml_amd.utils.atomic_io import atomic_write_json, atomic_write_text
+
+_SECTION_BUILDERS = [
+    ("system", system_section.build),
+    ("process", process_section.build),
+    ("step_time", step_time_section.build),
+    ("step_memory", step_memory_section.build),
+]
+
+
+def _run_duration_s(db_path: str) -> Optional[float]:
+    try:
+        conn = sqlite3.connect(f"file:{db_path}?mode=ro", uri=True)
+        try:
+            lo_hi = []
+            for table in ("step_time_samples", "system_samples", "process_samples"):
+                try:
+                    row = conn.execute(
+                        f"SELECT MIN(timestamp), MAX(timestamp) FROM {table}"
+                    ).fetchone()
+                    if row and row[0] is not None:
+                        lo_hi.append(row)
+                except sqlite3.Error:
+                    continue
+            if not lo_hi:
+                return None
+            lo = min(r[0] for r in lo_hi)
+            hi = max(r[1] for r in lo_hi)
+            return max(0.0, hi - lo)
+        finally:
+            conn.close()
+    except sqlite3.Error:
+        return None
+
+
+class FinalReportGenerator:
+    def __init__(self, db_path: str, run_name: Optional[str] = None) -> None:
+        self.db_path = db_path
+        self.run_name = run_name
+
+    def generate(self) -> dict:
+        sections = {}
+        for name, builder in _SECTION_BUILDERS:
+            try:
+                sections[name] = builder(self.db_path)
+            except Exception as exc:  # a broken section never kills the report
+                sections[name] = {
+                    "metadata": {"mode": "no_data", "error": repr(exc)},
+                    "diagnosis": {
+                        "kind": "SECTION_ERROR",
+                        "status": "SECTION ERROR",
+                        "severity": "info",
+                        "summary": f"Section failed to build: {exc!r}",
+                        "action": "",
+                    },
+                    "issues": [],
+                    "global": {},
+                    "groups": {"by": "global_rank", "rows": {}},
+                    "units": {},
+                    "card": "",
+                }
+
+        primary = build_primary_diagnosis(
+            sections.get("step_time", {}), sections.get("system", {})
+        )
+
+        st_md = sections.get("step_time", {}).get("metadata", {})
+        ranks_seen = st_md.get("global_ranks_seen") or []
+        world_size = None
+        rows = sections.get("step_time", {}).get("groups", {}).get("rows", {})
+        for row in rows.values():
+            ws = row.get("identity", {}).get("world_size")
+            if ws:
+                world_size = ws
+                break
+        payload = {
+            "schema_version": SCHEMA_VERSION,
+            "generated_at": datetime.datetime.now(datetime.timezone.utc).isoformat(),
+            "duration_s": _run_duration_s(self.db_path),
+            "meta": {
+                "run_name": self.run_name,
+                "mode": st_md.get("mode", "no_data"),
+                "world_size": world_size or (len(ranks_seen) or None),
+                "nodes_observed": st_md.get("nodes_observed"),
+                "gpus_observed": st_md.get("gpus_observed"),
+            },
+            "primary_diagnosis": primary,
+            **sections,
+        }
+        payload["text"] = build_verdict_text(payload)
+        return payload
+
+
+def build_verdict_text(payload: dict) -> str:
+    primary = payload.get("primary_diagnosis", {})
+    lines = [
+        "TraceML-AMD Verdict: "
+        + str(primary.get("status", "UNKNOWN"))
+        + f"  [{primary.get('severity', 'info')}]",
+        "Why:  " + str(primary.get("summary", "")),
+    ]
+    action = primary.get("action")
+    if action:
+        lines.append("Next: " + str(action))
+    lines.append("")
+    lines.append("Sections:")
+    for name in ("step_time", "step_memory", "system", "process"):
+        diag = payload.get(name, {}).get("diagnosis", {})
+        if diag:
+            lines.append(
+                f"  {name:<12} {diag.get('status', ''):<28} "
+                f"[{diag.get('severity', 'info')}]"
+            )
+    lines.append("")
+    for name in ("step_time", "step_memory", "system", "process"):
+        card = payload.get(name, {}).get("card")
+        if card:
+            lines.append(card)
+            lines.append("")
+    return "\n".join(lines).rstrip() + "\n"
+
+
+def write_summary_artifacts(
+    payload: dict, session_dir: str, html: bool = False
+) -> dict:
+    from traceml_amd.sdk import protocol
+
+    paths = {
+        "json": protocol.summary_json_path(session_dir),
+        "txt": protocol.summary_txt_path(session_dir),
+    }
+    atomic_write_json(paths["json"], payload)
+    atomic_write_text(paths["txt"], payload.get("text", ""))
+    if html:
+        from traceml_amd.reporting.html.document import render_html
+
+        paths["html"] = protocol.summary_html_path(session_dir)
+        atomic_write_text(paths["html"], render_html(payload))
+    return paths
+
+
+def generate_summary(
+    db_path: str,
+    session_dir: str,
+    run_name: Optional[str] = None,
+    html: bool = False,
+) -> dict:
+    payload = FinalReportGenerator(db_path, run_name=run_name).generate()
+    write_summary_artifacts(payload, session_dir, html=html)
+    return payload
