@@ -153,15 +153,22 @@ class TraceMLAggregator:
     def _settle_end_of_run_telemetry(self, budget_sec: float) -> None:
         """Keep ingesting until all expected ranks said rank_finished or the
         budget ends (reference: trace_aggregator.py:440-499)."""
-        deadline = time.time() + budget_sec
+        start = time.time()
+        deadline = start + budget_sec
         expected = self._expected_ranks()
+        #: with no expected-rank contract and nothing ever received, a short
+        #: grace beats burning the whole settle budget on an empty run
+        empty_grace_deadline = start + min(3.0, budget_sec)
         while time.time() < deadline:
             self.server.wait_for_data(timeout=0.25)
             self._drain_tcp()
             if expected is not None:
                 if len(self._finished_ranks) >= expected:
                     break
-            elif self._seen_ranks and self._finished_ranks >= self._seen_ranks:
+            elif self._seen_ranks:
+                if self._finished_ranks >= self._seen_ranks:
+                    break
+            elif time.time() >= empty_grace_deadline:
                 break
         self._drain_tcp()
 

@@ -1,0 +1,221 @@
+"""Launcher orchestration (reference: launcher/commands.py:276-820).
+
+``launch_process``: resolve config → set the TRACEML_* env contract →
+write manifests → [node 0] spawn the aggregator process → wait for its TCP
+listen → spawn torchrun with the executor → monitor → terminate the
+aggregator with the finalize budget → verify final_summary.json (summary
+mode hard-fails without it) → manifest status.
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import time
+from typing import List, Optional
+
+from traceml_amd.config.yaml_loader import resolve_config
+from traceml_amd.launcher import manifest as manifest_mod
+from traceml_amd.launcher.launch_config import (
+    AggregatorLaunchConfig,
+    RunIdentity,
+    TorchrunLaunchConfig,
+)
+from traceml_amd.launcher.process import (
+    spawn_process_group,
+    terminate_process_group,
+    wait_for_tcp_listen,
+)
+from traceml_amd.runtime.session import generate_session_id, session_dir
+from traceml_amd.runtime.settings import TraceMLSettings, apply_settings_to_env
+from traceml_amd.sdk import protocol
+
+
+def _executor_path() -> str:
+    from traceml_amd.runtime import executor
+
+    return os.path.abspath(executor.__file__)
+
+
+def launch_process(
+    script: str,
+    script_args: List[str],
+    nproc_per_node: int = 1,
+    nnodes: int = 1,
+    node_rank: int = 0,
+    master_addr: str = "127.0.0.1",
+    master_port: int = 29500,
+    cli_overrides: Optional[dict] = None,
+    echo: bool = True,
+) -> int:
+    settings = resolve_config(cli_overrides)
+    identity = RunIdentity(
+        run_name=settings.run_name, session_id=settings.session_id
+    )
+    identity.validate(nnodes)
+    if not settings.session_id:
+        settings.session_id = settings.run_name or generate_session_id()
+    settings.expected_ranks = nnodes * nproc_per_node
+
+    sdir = session_dir(settings.logs_dir, settings.session_id)
+    os.makedirs(sdir, exist_ok=True)
+    env = dict(os.environ)
+    apply_settings_to_env(settings, env)
+
+    manifest_mod.write_run_manifest(
+        sdir,
+        manifest_mod.STATUS_STARTING,
+        script=script,
+        script_args=script_args,
+        world_size=nnodes * nproc_per_node,
+        nnodes=nnodes,
+        run_name=settings.run_name,
+    )
+    manifest_mod.write_code_manifest(sdir, script)
+
+    agg_config = AggregatorLaunchConfig(node_rank=node_rank)
+    agg_proc = None
+    telemetry_status = "ok"
+    try:
+        if agg_config.is_owner:
+            agg_proc, _ = spawn_process_group(
+                [sys.executable, "-m", "traceml_amd.aggregator.main"], env=env
+            )
+            if not wait_for_tcp_listen(
+                settings.aggregator_host,
+                settings.aggregator_port,
+                timeout=30.0,
+                proc=agg_proc,
+            ):
+                print(
+                    "[TraceML-AMD] aggregator did not start listening; "
+                    "continuing without telemetry",
+                    file=sys.stderr,
+                )
+                telemetry_status = "degraded"
+
+        torchrun = TorchrunLaunchConfig(
+            nproc_per_node=nproc_per_node,
+            nnodes=nnodes,
+            node_rank=node_rank,
+            master_addr=master_addr,
+            master_port=master_port,
+            script=script,
+            script_args=script_args,
+        )
+        manifest_mod.update_status(sdir, manifest_mod.STATUS_RUNNING)
+        train_proc, _ = spawn_process_group(
+            torchrun.to_command(_executor_path()), env=env
+        )
+
+        # Monitor loop: if the aggregator dies early, training continues
+        # with degraded telemetry (fail-open; reference commands.py:585-600).
+        while True:
+            code = train_proc.poll()
+            if code is not None:
+                break
+            if agg_proc is not None and agg_proc.poll() is not None:
+                telemetry_status = "degraded"
+                agg_proc = None
+            time.sleep(0.5)
+        train_code = train_proc.returncode
+    finally:
+        if agg_proc is not None:
+            terminate_process_group(
+                agg_proc, grace_sec=float(settings.finalize_timeout)
+            )
+
+    summary_path = protocol.summary_json_path(sdir)
+    summary_ok = os.path.exists(summary_path)
+    status = (
+        manifest_mod.STATUS_COMPLETED
+        if train_code == 0
+        else manifest_mod.STATUS_FAILED
+    )
+    manifest_mod.update_status(
+        sdir,
+        status,
+        extra={
+            "exit_code": train_code,
+            "telemetry_status": telemetry_status,
+            "artifacts": manifest_mod.collect_existing_artifacts(sdir),
+        },
+    )
+    if echo:
+        if summary_ok:
+            print(f"[TraceML-AMD] final summary: {summary_path}")
+            try:
+                with open(protocol.summary_txt_path(sdir), "r", encoding="utf-8") as f:
+                    print(f.read())
+            except OSError:
+                pass
+    if settings.mode == "summary" and not summary_ok and telemetry_status == "ok":
+        print(
+            "[TraceML-AMD] ERROR: summary mode but no final_summary.json was "
+            "produced",
+            file=sys.stderr,
+        )
+        return train_code if train_code != 0 else 3
+    return train_code
+
+
+def run_serve(cli_overrides: Optional[dict] = None) -> int:
+    """Standalone aggregator for traceml_amd.init() direct launches
+    (reference: commands.py:702)."""
+    settings = resolve_config(cli_overrides)
+    if not settings.session_id:
+        settings.session_id = settings.run_name or generate_session_id()
+    apply_settings_to_env(settings)
+    from traceml_amd.aggregator.main import main as aggregator_main
+
+    return aggregator_main()
+
+
+def run_inspect(path: str) -> int:
+    """Dump per-rank msgpack backups (reference: commands.py:733)."""
+    import json
+
+    from traceml_amd.database.writer import read_msgpack_table
+
+    if os.path.isfile(path):
+        files = [path]
+    else:
+        files = []
+        for root, _dirs, names in os.walk(path):
+            files.extend(
+                os.path.join(root, n) for n in names if n.endswith(".msgpack")
+            )
+    for f in sorted(files):
+        rows = read_msgpack_table(f)
+        print(f"== {f} ({len(rows)} rows)")
+        for row in rows[:20]:
+            print(json.dumps(row, default=str))
+        if len(rows) > 20:
+            print(f"... {len(rows) - 20} more")
+    return 0
+
+
+def run_view(path: str) -> int:
+    """Re-print a saved final summary (reference: reporting/view/command.py:41)."""
+    import json
+
+    try:
+        with open(path, "r", encoding="utf-8") as f:
+            payload = json.load(f)
+    except (OSError, ValueError) as exc:
+        print(f"cannot read {path}: {exc}", file=sys.stderr)
+        return 1
+    text = payload.get("text")
+    if not text:
+        from traceml_amd.reporting.final import build_verdict_text
+
+        text = build_verdict_text(payload)
+    print(text)
+    return 0
+
+
+def run_compare(path_a: str, path_b: str) -> int:
+    from traceml_amd.reporting.compare.command import compare_files
+
+    return compare_files(path_a, path_b)
