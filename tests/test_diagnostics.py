@@ -1,0 +1,157 @@
+"""Diagnosis classification tests over the canonical SQLite scenarios —
+the cross-surface contract suite (mirrors reference
+tests/step_time/test_contract_baseline.py + tests/diagnostics/*)."""
+
+import os
+
+import pytest
+
+from tests import scenarios
+from traceml_amd.steptime.pipeline import StepTimePipeline
+
+
+@pytest.fixture
+def db_path(tmp_path):
+    return str(tmp_path / "telemetry.sqlite")
+
+
+def run_pipeline(db_path):
+    return StepTimePipeline(db_path, profile="summary").run()
+
+
+def test_healthy_is_balanced_or_compute(db_path):
+    scenarios.healthy_ddp(ranks=4, steps=30).write(db_path)
+    result = run_pipeline(db_path)
+    assert result.diagnosis.primary.kind in ("BALANCED", "COMPUTE_BOUND")
+
+
+def test_input_bound_critical(db_path):
+    scenarios.input_bound(steps=30).write(db_path)
+    result = run_pipeline(db_path)
+    primary = result.diagnosis.primary
+    assert primary.kind == "INPUT_BOUND"
+    assert primary.severity == "crit"  # 64% share, confident window
+    assert primary.score is not None and primary.score > 0.5
+
+
+def test_input_bound_warn_below_confident_window(db_path):
+    scenarios.input_bound(steps=5).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "INPUT_BOUND"
+    assert primary.severity == "warn"  # crit requires >= 20 steps
+
+
+def test_input_straggler_culprit_is_lowest_visible(db_path):
+    scenarios.input_straggler(ranks=4, steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "INPUT_STRAGGLER"
+    assert primary.ranks == [2]  # the slow-dataloader rank, NOT a waiter
+    assert primary.severity == "crit"
+    assert primary.evidence["type"] == "rank_comparison"
+    # MI355X extra: measured ddp_comm corroboration present
+    assert "ddp_comm_ms_per_rank" in primary.evidence
+
+
+def test_compute_straggler(db_path):
+    scenarios.compute_straggler(ranks=4, steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "COMPUTE_STRAGGLER"
+    assert primary.ranks == [1]
+
+
+def test_residual_heavy(db_path):
+    scenarios.residual_heavy(steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "RESIDUAL_HEAVY"
+
+
+def test_warmup_single_step(db_path):
+    scenarios.healthy_ddp(ranks=1, steps=1).write(db_path)
+    assert run_pipeline(db_path).diagnosis.primary.kind == "WARMUP"
+
+
+def test_no_data(db_path):
+    scenarios.StepTimeScenario("empty", {}, steps=0).write(db_path)
+    assert run_pipeline(db_path).diagnosis.primary.kind == "NO_DATA"
+
+
+def test_fsdp_straggler_capped_at_warn(db_path):
+    scenario = scenarios.compute_straggler(ranks=4, steps=30)
+    scenario.strategy = "fsdp"
+    scenario.write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    if "STRAGGLER" in primary.kind:
+        assert primary.severity == "warn"
+
+
+def test_issues_first_is_diagnosis_invariant(db_path):
+    scenarios.input_bound(steps=30).write(db_path)
+    result = run_pipeline(db_path)
+    payload = result.diagnosis.to_payload()
+    assert payload["issues"][0] == payload["diagnosis"]
+    assert len(payload["issues"]) >= 1
+
+
+# -- step-memory diagnosis ---------------------------------------------------
+
+
+def test_memory_pressure_crit(db_path):
+    gib = 1 << 30
+    scenarios.write_memory_rows(
+        db_path, {0: (270 * gib, 283 * gib)}, capacity=288 * gib
+    )
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+
+    result = diagnose_step_memory(load_memory_series(db_path))
+    assert result.primary.kind == "HIGH_MEMORY_PRESSURE"
+    assert result.primary.severity == "crit"  # 98% of capacity
+
+
+def test_memory_creep_confirmed(db_path):
+    gib = 1 << 30
+    scenarios.write_memory_rows(
+        db_path,
+        {0: (10 * gib, 12 * gib)},
+        capacity=288 * gib,
+        steps=1000,
+        creep_bytes_per_step=2 * (1 << 20),  # 2 MiB/step -> ~2 GiB over run
+    )
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+
+    result = diagnose_step_memory(load_memory_series(db_path))
+    kinds = [i.kind for i in result.issues]
+    assert "MEMORY_CREEP_CONFIRMED" in kinds
+
+
+def test_memory_normal(db_path):
+    gib = 1 << 30
+    scenarios.write_memory_rows(db_path, {0: (40 * gib, 48 * gib), 1: (40 * gib, 48 * gib)})
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+
+    result = diagnose_step_memory(load_memory_series(db_path))
+    assert result.primary.kind == "NORMAL"
+
+
+def test_rank_memory_imbalance(db_path):
+    gib = 1 << 30
+    scenarios.write_memory_rows(
+        db_path, {0: (200 * gib, 220 * gib), 1: (90 * gib, 100 * gib)},
+        capacity=288 * gib,
+    )
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+
+    result = diagnose_step_memory(load_memory_series(db_path))
+    kinds = [i.kind for i in result.issues]
+    assert "RANK_MEMORY_IMBALANCE" in kinds

@@ -1,0 +1,114 @@
+"""Final-summary contract tests: section shape, nullability, primary
+promotion, text report, HTML, compare
+(mirrors reference tests/reporting/summary/*)."""
+
+import json
+
+import pytest
+
+from tests import scenarios
+from traceml_amd.reporting.final import FinalReportGenerator, generate_summary
+from traceml_amd.steptime.model import STEP_TIME_METRIC_NAMES
+
+
+@pytest.fixture
+def db_path(tmp_path):
+    return str(tmp_path / "telemetry.sqlite")
+
+
+def test_full_payload_shape(db_path, tmp_path):
+    scenarios.input_bound(steps=30).write(db_path)
+    payload = generate_summary(db_path, str(tmp_path), html=True)
+
+    assert payload["schema_version"] == 1.7
+    for key in ("generated_at", "duration_s", "meta", "primary_diagnosis",
+                "system", "process", "step_time", "step_memory", "text"):
+        assert key in payload
+
+    st = payload["step_time"]
+    for key in ("metadata", "diagnosis", "issues", "global", "groups", "units",
+                "card"):
+        assert key in st
+    assert st["issues"][0] == st["diagnosis"]
+    assert st["metadata"]["section_metric_names"] == STEP_TIME_METRIC_NAMES
+    assert st["global"]["index_by"] == st["groups"]["by"] == "global_rank"
+    assert st["global"]["window"]["alignment"] == "common_steps"
+
+    # nullability: every metric key present; unmeasured are null not 0
+    row = st["groups"]["rows"]["0"]["metrics"]
+    assert set(row) == set(STEP_TIME_METRIC_NAMES)
+    assert row["ddp_comm_ms"] is None  # not measured in this scenario
+    assert st["global"]["average"]["ddp_comm_ms"] is None
+    assert row["input_wait_ms"] == pytest.approx(128.0)
+
+    # artifacts on disk
+    assert (tmp_path / "final_summary.json").exists()
+    assert (tmp_path / "final_summary.txt").exists()
+    assert (tmp_path / "final_summary.html").exists()
+    text = (tmp_path / "final_summary.txt").read_text()
+    assert text.startswith("TraceML-AMD Verdict")
+    assert "INPUT-BOUND" in text
+    html = (tmp_path / "final_summary.html").read_text()
+    assert "<svg" in html and "INPUT" in html
+
+
+def test_primary_promotion_straggler(db_path):
+    scenarios.input_straggler(steps=30).write(db_path)
+    payload = FinalReportGenerator(db_path).generate()
+    primary = payload["primary_diagnosis"]
+    assert primary["kind"] == "INPUT_STRAGGLER"
+    assert primary["section"] == "step_time"
+    assert primary["scope"] == "performance"
+    assert payload["meta"]["world_size"] == 4
+
+
+def test_primary_insufficient_data(db_path):
+    scenarios.StepTimeScenario("empty", {}, steps=0).write(db_path)
+    payload = FinalReportGenerator(db_path).generate()
+    assert payload["primary_diagnosis"]["kind"] == "INSUFFICIENT_STEP_TIME_DATA"
+    assert payload["primary_diagnosis"]["evidence"]["type"] == "insufficient_data"
+
+
+def test_generator_survives_missing_db(tmp_path):
+    payload = FinalReportGenerator(str(tmp_path / "missing.sqlite")).generate()
+    assert payload["primary_diagnosis"]["kind"] == "INSUFFICIENT_STEP_TIME_DATA"
+
+
+def test_compare_detects_improvement(db_path, tmp_path):
+    from traceml_amd.reporting.compare.command import compare_payloads
+
+    scenarios.input_bound(steps=30).write(db_path)
+    slow = FinalReportGenerator(db_path).generate()
+    fast_db = str(tmp_path / "fast.sqlite")
+    scenarios.healthy_ddp(ranks=1, steps=30).write(fast_db)
+    fast = FinalReportGenerator(fast_db).generate()
+
+    result = compare_payloads(slow, fast)
+    assert result["verdict"] == "IMPROVEMENT"
+    step = next(m for m in result["metrics"] if m["metric"] == "step_time_ms")
+    assert step["status"] == "IMPROVEMENT"
+    reverse = compare_payloads(fast, slow)
+    assert reverse["verdict"] == "REGRESSION"
+
+
+def test_compare_neutral_on_identical(db_path):
+    from traceml_amd.reporting.compare.command import compare_payloads
+
+    scenarios.healthy_ddp(ranks=2, steps=30).write(db_path)
+    a = FinalReportGenerator(db_path).generate()
+    result = compare_payloads(a, json.loads(json.dumps(a)))
+    assert result["verdict"] in ("NEUTRAL", "INCOMPARABLE")
+
+
+def test_summary_projection_flat_dict(db_path):
+    from traceml_amd.sdk.summary_client import compact_summary
+
+    scenarios.input_bound(steps=30).write(db_path)
+    payload = FinalReportGenerator(db_path).generate()
+    flat = compact_summary(payload)
+    assert flat["traceml/verdict_kind"] == "INPUT_BOUND"
+    assert isinstance(flat["traceml/step_time/step_time_ms"], float)
+    # tracker-friendly: values are scalars only
+    assert all(
+        isinstance(v, (int, float, str, type(None))) for v in flat.values()
+    )

@@ -1,0 +1,86 @@
+"""Process-level smoke test: the real launcher + aggregator + executor run an
+embedded TinyMLP CPU script and final_summary.json appears with the right
+shape (mirrors reference tests/runtime/test_final_summary_smoke.py:25-80)."""
+
+import json
+import os
+import subprocess
+import sys
+import textwrap
+
+import pytest
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+SCRIPT = textwrap.dedent(
+    """
+    import torch, torch.nn as nn
+    from torch.utils.data import DataLoader, TensorDataset
+    import traceml_amd
+
+    model = nn.Sequential(nn.Linear(32, 64), nn.ReLU(), nn.Linear(64, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    ds = TensorDataset(torch.randn(320, 32), torch.randn(320, 4))
+    dl = DataLoader(ds, batch_size=8)
+    n = 0
+    for epoch in range(2):
+        for x, y in dl:
+            if n >= 60:
+                break
+            with traceml_amd.trace_step(model):
+                opt.zero_grad()
+                ((model(x) - y) ** 2).mean().backward()
+                opt.step()
+            n += 1
+    print("steps:", n)
+    """
+)
+
+
+@pytest.mark.timeout(240)
+def test_final_summary_smoke(tmp_path):
+    script = tmp_path / "train_tiny.py"
+    script.write_text(SCRIPT)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "60"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "traceml_amd",
+            "run",
+            "--logs-dir",
+            str(tmp_path / "logs"),
+            "--session-id",
+            "smoke",
+            "--aggregator-port",
+            "29877",
+            "--master-port",
+            "29601",
+            str(script),
+        ],
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=220,
+        cwd=REPO_ROOT,
+    )
+    session = tmp_path / "logs" / "smoke"
+    summary_path = session / "final_summary.json"
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    assert summary_path.exists(), (proc.stdout[-2000:], proc.stderr[-2000:])
+
+    payload = json.loads(summary_path.read_text())
+    assert payload["schema_version"] == 1.7
+    assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 60
+    assert payload["primary_diagnosis"]["kind"]
+    assert (session / "final_summary.txt").exists()
+    assert (session / "manifest.json").exists()
+    manifest = json.loads((session / "manifest.json").read_text())
+    assert manifest["status"] == "completed"
+    # per-rank msgpack backups exist
+    rank_data = session / "r0" / "data"
+    assert rank_data.is_dir()
+    assert any(rank_data.rglob("*.msgpack"))
