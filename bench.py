@@ -159,13 +159,19 @@ def main():
     rank, local_rank, world_size, use_gpu = setup_distributed(args)
     device = torch.device(f"cuda:{local_rank}" if use_gpu else "cpu")
 
+    if use_gpu:
+        # MIOpen find-mode kernel selection: tune once in warmup, then stable
+        torch.backends.cudnn.benchmark = True
+
     model, optimizer, loader, loss_fn, batch = build_workload(args, device, use_gpu)
     if world_size > 1:
         from torch.nn.parallel import DistributedDataParallel as DDP
 
         model = DDP(model, device_ids=[local_rank] if use_gpu else None)
 
-    # ---- phase 1: untraced baseline ----
+    # ---- phase 1: untraced baseline (first pass absorbs cold-GPU effects:
+    # MIOpen find, allocator growth, DVFS ramp; it is re-measured
+    # interleaved with the traced phase below and the MIN is used) ----
     t_off = run_phase(
         model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
         traced=False,
@@ -210,14 +216,24 @@ def main():
         attach_ddp_comm_timing(model)
         enable_rank_stats_exchange()
 
-    # ---- phase 2: traced ----
+    # ---- interleaved measurement: on, off, on — overhead compares the
+    # best traced pass against the best untraced pass at equal thermal /
+    # clock / allocator state (a fixed order reads DVFS ramp as overhead) ----
     t_on = run_phase(
         model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
         traced=True, trace_ctx=trace_step,
     )
+    t_off2 = run_phase(
+        model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
+        traced=False,
+    )
+    t_on2 = run_phase(
+        model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
+        traced=True, trace_ctx=trace_step,
+    )
 
-    t_off = max_over_ranks(t_off, world_size, use_gpu)
-    t_on = max_over_ranks(t_on, world_size, use_gpu)
+    t_off = max_over_ranks(min(t_off, t_off2), world_size, use_gpu)
+    t_on = max_over_ranks(min(t_on, t_on2), world_size, use_gpu)
 
     # ---- teardown + diagnosis (outside the timed regions) ----
     handle.stop()

@@ -250,6 +250,10 @@ def reset_for_tests() -> None:
         timing.clear_for_tests()
         step_memory.clear_for_tests()
         environment.reset_for_tests()
+        from traceml_amd.sdk import instrumentation as _sdk_instr
+
+        _sdk_instr._cached_identity = None
+        _sdk_instr._cached_mem_tracker = None
         state.session_state().reset_for_tests()
         state.recording_state().reset_for_tests()
         _active_config = None

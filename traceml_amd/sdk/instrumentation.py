@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import functools
 import time
-from contextlib import contextmanager
+
 
 from traceml_amd.core import event_names
 from traceml_amd.core.arming import is_tracing_armed, phase_flags
@@ -30,8 +30,89 @@ from traceml_amd.runtime import environment, state
 from traceml_amd.runtime.identity import resolve_runtime_identity
 from traceml_amd.sdk import initial
 
+# Hot-path caches: identity, the memory tracker and the forward-target set
+# are process-stable; the per-step bracket must stay at a few microseconds
+# of Python (class-based context manager — the generator form costs ~2x).
+_cached_identity = None
+_cached_mem_tracker = None
+_cached_targets: dict = {}
 
-@contextmanager
+
+class _NoopStep:
+    __slots__ = ("_advance",)
+
+    def __init__(self, advance: bool) -> None:
+        self._advance = advance
+
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *exc):
+        if self._advance:
+            state.session_state().advance()
+        return False
+
+
+class _TraceStep:
+    __slots__ = ("_model", "_config", "_recording", "_mem", "_event")
+
+    def __init__(self, model, config, recording) -> None:
+        self._model = model
+        self._config = config
+        self._recording = recording
+
+    def __enter__(self):
+        global _cached_identity, _cached_mem_tracker
+        model = self._model
+        config = self._config
+        if _cached_identity is None:
+            _cached_identity = resolve_runtime_identity()
+        environment.publish_runtime_environment_once(_cached_identity, model)
+        if _cached_mem_tracker is None:
+            _cached_mem_tracker = StepMemoryTracker(model)
+        self._mem = _cached_mem_tracker
+        self._mem.reset()
+
+        flags = phase_flags()
+        flags.in_step = True
+        if model is not None:
+            key = id(model)
+            targets = _cached_targets.get(key)
+            if targets is None:
+                targets = forward_target_ids(model)
+                if len(_cached_targets) > 64:
+                    _cached_targets.clear()
+                _cached_targets[key] = targets
+            flags.forward_targets = targets
+            flags.forward_enabled = config.patch_forward
+        else:
+            flags.forward_targets = ()
+            flags.forward_enabled = False
+        flags.backward_enabled = config.patch_backward
+        flags.h2d_enabled = config.patch_h2d
+        flags.optimizer_enabled = config.mode == "auto"
+        if config.auto_optimizer_hooks:
+            ensure_optimizer_timing_installed()
+        self._event = open_event(event_names.STEP_TIME)
+        return None
+
+    def __exit__(self, *exc):
+        close_event(self._event)
+        flags = phase_flags()
+        flags.in_step = False
+        flags.forward_enabled = False
+        flags.backward_enabled = False
+        flags.h2d_enabled = False
+        flags.optimizer_enabled = False
+        flags.forward_targets = ()
+        step = state.session_state().advance()
+        self._mem.record(step)
+        flush_step_events(step)
+        self._recording.mark_trace_step_flushed()
+        _kick_rank_stats(step)
+        return False
+
+
 def trace_step(model=None):
     config = initial.get_active_config()
     recording = state.recording_state()
@@ -41,43 +122,8 @@ def trace_step(model=None):
         or not is_tracing_armed()
         or not recording.should_record_trace_events()
     ):
-        yield
-        if config is not None and not config.noop:
-            state.session_state().advance()
-        return
-
-    identity = resolve_runtime_identity()
-    environment.publish_runtime_environment_once(identity, model)
-
-    mem_tracker = StepMemoryTracker(model)
-    mem_tracker.reset()
-
-    flags = phase_flags()
-    flags.in_step = True
-    flags.forward_targets = forward_target_ids(model) if model is not None else ()
-    flags.forward_enabled = config.patch_forward and model is not None
-    flags.backward_enabled = config.patch_backward
-    flags.h2d_enabled = config.patch_h2d
-    flags.optimizer_enabled = config.mode == "auto"
-    if config.auto_optimizer_hooks:
-        ensure_optimizer_timing_installed()
-
-    step_event = open_event(event_names.STEP_TIME)
-    try:
-        yield
-    finally:
-        close_event(step_event)
-        flags.in_step = False
-        flags.forward_enabled = False
-        flags.backward_enabled = False
-        flags.h2d_enabled = False
-        flags.optimizer_enabled = False
-        flags.forward_targets = ()
-        step = state.session_state().advance()
-        mem_tracker.record(step)
-        flush_step_events(step)
-        recording.mark_trace_step_flushed()
-        _kick_rank_stats(step)
+        return _NoopStep(advance=config is not None and not config.noop)
+    return _TraceStep(model, config, recording)
 
 
 def _kick_rank_stats(step: int) -> None:
