@@ -26,10 +26,11 @@ def _clean_traceml_state():
         del os.environ[k]
     yield
     from traceml_amd.sdk import initial
-    from traceml_amd.runtime import session
+    from traceml_amd.runtime import session, stdout_capture
 
     initial.reset_for_tests()
     session.reset_for_tests()
+    stdout_capture.reset_for_tests()
     for k in [k for k in os.environ if k.startswith("TRACEML_")]:
         del os.environ[k]
     os.environ.update(saved_env)

@@ -1,0 +1,165 @@
+"""Integration conformance — the StreamContract gate: each integration
+DECLARES the step-time streams it owes, and a tiny CPU run under its
+documented init path must emit >= 1 event per declared stream
+(mirrors reference tests/integrations/test_telemetry_conformance.py)."""
+
+import pytest
+import torch
+import torch.nn as nn
+
+from traceml_amd.core import event_names, timing
+from traceml_amd.integrations._capability import REQUIRED_STREAMS
+
+_STREAM_TO_EVENT = {
+    "forward_time": event_names.FORWARD,
+    "backward_time": event_names.BACKWARD,
+    "optimizer_step": event_names.OPTIMIZER,
+    "step_time": event_names.STEP_TIME,
+    "h2d_time": event_names.H2D,
+    "dataloader_next": event_names.DATALOADER,
+}
+
+
+def _collected_event_names():
+    batches = timing.drain_step_time_queue()
+    return {e.name for b in batches for e in b.events}
+
+
+def test_required_streams_registry_covers_known_integrations():
+    assert set(REQUIRED_STREAMS) == {"huggingface", "lightning", "accelerate", "ray"}
+    for streams in REQUIRED_STREAMS.values():
+        for s in streams:
+            assert s in _STREAM_TO_EVENT
+
+
+def test_capability_warning_on_disabled_stream(capsys):
+    from traceml_amd.integrations._capability import warn_if_missing_streams
+    from traceml_amd.runtime.settings import TraceMLSettings
+    from traceml_amd.sdk.initial import TraceMLInitConfig
+
+    config = TraceMLInitConfig(
+        mode="custom", patch_forward=False, settings=TraceMLSettings()
+    )
+    missing = warn_if_missing_streams("huggingface", config)
+    assert missing == ["forward_time"]
+    assert "forward_time" in capsys.readouterr().err
+
+
+class _HFCompatModel(nn.Module):
+    """Minimal model with the HF Trainer contract (returns dict with loss)."""
+
+    def __init__(self):
+        super().__init__()
+        self.net = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 2))
+
+    def forward(self, x=None, labels=None):
+        logits = self.net(x)
+        loss = nn.functional.cross_entropy(logits, labels)
+        return {"loss": loss, "logits": logits}
+
+
+@pytest.mark.timeout(300)
+def test_hf_trainer_callback_emits_owed_streams(tmp_path, armed_auto_config):
+    transformers = pytest.importorskip("transformers")
+    from transformers import Trainer, TrainingArguments
+
+    from traceml_amd.integrations.huggingface import TraceMLTrainerCallback
+
+    ds = [
+        {"x": torch.randn(8), "labels": torch.tensor(i % 2)} for i in range(16)
+    ]
+    args = TrainingArguments(
+        output_dir=str(tmp_path),
+        per_device_train_batch_size=4,
+        num_train_epochs=1,
+        logging_strategy="no",
+        save_strategy="no",
+        report_to=[],
+        use_cpu=True,
+        disable_tqdm=True,
+    )
+    trainer = Trainer(
+        model=_HFCompatModel(),
+        args=args,
+        train_dataset=ds,
+        callbacks=[TraceMLTrainerCallback()],
+    )
+    trainer.train()
+    names = _collected_event_names()
+    for stream in REQUIRED_STREAMS["huggingface"]:
+        assert _STREAM_TO_EVENT[stream] in names, f"HF lost stream {stream}"
+
+
+@pytest.mark.timeout(120)
+def test_hf_grad_accumulation_folds_micro_batches(tmp_path, armed_auto_config):
+    pytest.importorskip("transformers")
+    from transformers import Trainer, TrainingArguments
+
+    from traceml_amd.integrations.huggingface import TraceMLTrainerCallback
+    from traceml_amd.runtime import state
+
+    start_step = state.session_state().current_step
+    ds = [
+        {"x": torch.randn(8), "labels": torch.tensor(i % 2)} for i in range(16)
+    ]
+    args = TrainingArguments(
+        output_dir=str(tmp_path),
+        per_device_train_batch_size=2,
+        gradient_accumulation_steps=4,  # 16 samples -> 2 optimizer steps
+        num_train_epochs=1,
+        logging_strategy="no",
+        save_strategy="no",
+        report_to=[],
+        use_cpu=True,
+        disable_tqdm=True,
+    )
+    trainer = Trainer(
+        model=_HFCompatModel(), args=args, train_dataset=ds,
+        callbacks=[TraceMLTrainerCallback()],
+    )
+    trainer.train()
+    assert state.session_state().current_step - start_step == 2
+
+
+def test_lightning_callback_manual_phases(armed_auto_config):
+    """Without Lightning installed, exercise the callback protocol directly
+    (the phase-timing logic is framework-independent)."""
+    from traceml_amd.integrations.lightning import TraceMLCallback
+
+    cb = TraceMLCallback()
+    model = nn.Linear(8, 2)
+    cb.on_train_batch_start(None, model, None, 0)
+    out = model(torch.randn(4, 8))  # timed: wrapped forward
+    cb.on_before_backward(None, model, out.sum())
+    out.sum().backward()
+    cb.on_after_backward(None, model)
+    cb.on_before_optimizer_step(None, model, None)
+    cb.on_before_zero_grad(None, model, None)
+    cb.on_train_batch_end(None, model, None, None, 0)
+
+    names = _collected_event_names()
+    assert event_names.STEP_TIME in names
+    assert event_names.FORWARD in names
+    assert event_names.BACKWARD in names
+    assert event_names.OPTIMIZER in names
+
+
+def test_accelerate_bracket(armed_auto_config):
+    from traceml_amd.integrations import accelerate as acc
+
+    model = nn.Linear(8, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    with acc.trace_step(model):
+        opt.zero_grad()
+        model(torch.randn(4, 8)).sum().backward()
+        opt.step()
+    names = _collected_event_names()
+    for stream in REQUIRED_STREAMS["accelerate"]:
+        assert _STREAM_TO_EVENT[stream] in names
+
+
+def test_ray_import_guard():
+    from traceml_amd.integrations import ray as ray_integration
+
+    with pytest.raises(ImportError):
+        ray_integration.TraceMLTorchTrainer(lambda cfg: None)

@@ -57,6 +57,22 @@ def main(argv=None) -> int:
     script, script_args = argv[0], argv[1:]
 
     settings = TraceMLSettings.from_env()
+
+    # Per-rank stdout/stderr capture: tee to session/r<k>/stdout_stderr.log
+    # and (rank 0, cli modes) to the wire via the stdout_stderr sampler.
+    try:
+        from traceml_amd.runtime.stdout_capture import install_stream_capture
+
+        identity = resolve_runtime_identity()
+        sdir = session_dir(settings.logs_dir, get_session_id(settings.session_id))
+        install_stream_capture(
+            os.path.join(
+                sdir, rank_dir_name(identity.local_rank), "stdout_stderr.log"
+            )
+        )
+    except Exception:
+        pass
+
     handle = None
     try:
         handle = lifecycle.start_runtime(settings, fail_open=True)

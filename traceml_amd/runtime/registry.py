@@ -58,6 +58,12 @@ def _rank_stats(db):
     return RankStatsSampler(db)
 
 
+def _stdout_stderr(db):
+    from traceml_amd.samplers.stdout_stderr import StdoutStderrSampler
+
+    return StdoutStderrSampler(db)
+
+
 DEFAULT_SAMPLER_REGISTRY: List[SamplerSpec] = [
     SamplerSpec("system", _system, rank_zero_only=True),
     SamplerSpec("runtime_environment", _runtime_env),
@@ -65,6 +71,10 @@ DEFAULT_SAMPLER_REGISTRY: List[SamplerSpec] = [
     SamplerSpec("step_time", _step_time),
     SamplerSpec("step_memory", _step_memory),
     SamplerSpec("rank_stats", _rank_stats),
+    SamplerSpec(
+        "stdout_stderr", _stdout_stderr, rank_zero_only=True,
+        modes=("cli", "dashboard"),
+    ),
 ]
 
 
