@@ -8,6 +8,11 @@ On a CPU box the demos run on gloo; on MI355X GPUs they run on RCCL.
 
 from __future__ import annotations
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))))
+
 import os
 import time
 

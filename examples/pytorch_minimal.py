@@ -4,6 +4,12 @@ Run:  traceml-amd run examples/pytorch_minimal.py
   or: python examples/pytorch_minimal.py   (after `traceml-amd serve` or standalone)
 """
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
 import torch
 import torch.nn as nn
 from torch.utils.data import DataLoader, TensorDataset

@@ -18,6 +18,7 @@ SCRIPT = textwrap.dedent(
     from torch.utils.data import DataLoader, TensorDataset
     import traceml_amd
 
+    traceml_amd.init()
     model = nn.Sequential(nn.Linear(32, 64), nn.ReLU(), nn.Linear(64, 4))
     opt = torch.optim.SGD(model.parameters(), lr=0.01)
     ds = TensorDataset(torch.randn(320, 32), torch.randn(320, 4))

@@ -49,11 +49,15 @@ class TraceMLFinalizationError(RuntimeError):
     pass
 
 
-def _build_display(mode: str) -> DisplayDriver:
+def _build_display(mode: str, dashboard_port: int = 8765) -> DisplayDriver:
     if mode == "cli":
         from traceml_amd.aggregator.display.cli import CLIDisplayDriver
 
         return CLIDisplayDriver()
+    if mode == "dashboard":
+        from traceml_amd.aggregator.display.dashboard import DashboardDisplayDriver
+
+        return DashboardDisplayDriver(port=dashboard_port)
     return SummaryDisplayDriver()
 
 
@@ -68,7 +72,9 @@ class TraceMLAggregator:
             port=self.settings.aggregator_port,
         )
         self.sqlite = SQLiteWriterSimple(self.db_path)
-        self.display = _build_display(self.settings.mode)
+        self.display = _build_display(
+            self.settings.mode, self.settings.dashboard_port
+        )
         self.summary_service = FinalSummaryService(
             self.session_dir,
             self.db_path,

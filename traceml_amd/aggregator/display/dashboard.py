@@ -1,0 +1,176 @@
+"""Web dashboard display driver (reference: NiceGUI driver +
+nicegui_sections, display_drivers/nicegui.py:535).
+
+The image has no NiceGUI, so this is a self-contained FastAPI/uvicorn app:
+``/`` serves a single-page dashboard that polls ``/api/live`` (the same
+step-time pipeline + section loaders every other surface uses) and renders
+the verdict, per-rank phase table, phase-share bar, memory and node health.
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Optional
+
+from traceml_amd.aggregator.display.base import DisplayDriver
+
+logger = logging.getLogger(__name__)
+
+_PAGE = """<!DOCTYPE html><html><head><meta charset="utf-8">
+<title>traceml-amd dashboard</title>
+<style>
+body{font-family:-apple-system,'Segoe UI',Roboto,sans-serif;background:#111;
+     color:#eee;margin:1.5rem auto;max-width:1100px}
+h1{font-size:1.2rem;color:#7ab8ff}
+.verdict{padding:.8rem 1rem;border-radius:8px;font-weight:600;margin:.6rem 0}
+.crit{background:#7a1f1f}.warn{background:#7a5a1f}.info{background:#1f4b7a}
+table{border-collapse:collapse;width:100%;font-size:.85rem;margin:.6rem 0}
+th,td{border:1px solid #333;padding:4px 8px;text-align:right}
+th:first-child,td:first-child{text-align:left}
+.bar{display:flex;height:26px;border-radius:4px;overflow:hidden;margin:.4rem 0}
+.bar div{height:100%}
+.legend{font-size:.75rem;color:#aaa}
+h2{font-size:1rem;color:#9ad;margin-top:1.4rem}
+.dim{color:#888;font-size:.75rem}
+</style></head><body>
+<h1>traceml-amd <span class="dim">MI355X training-step profiler</span></h1>
+<div id="content">loading…</div>
+<script>
+const COLORS={input:'#e07b39',h2d:'#8e44ad',forward:'#2d7dd2',
+  backward:'#1b998b',optimizer:'#97cc04',ddp_comm:'#d05ce3',residual:'#777'};
+async function tick(){
+  try{
+    const r=await fetch('/api/live');const d=await r.json();
+    let html='';
+    const diag=d.step_time.diagnosis||{};
+    html+=`<div class="verdict ${diag.severity||'info'}">${diag.status||'…'} — ${diag.summary||''}</div>`;
+    const shares=d.step_time.shares||{};
+    let bar='<div class="bar">';let legend='';
+    for(const [k,v] of Object.entries(shares)){
+      if(v&&COLORS[k]){bar+=`<div style="width:${Math.min(100,v*100)}%;background:${COLORS[k]}" title="${k} ${(v*100).toFixed(1)}%"></div>`;
+        legend+=`<span style="color:${COLORS[k]}">■</span> ${k} ${(v*100).toFixed(0)}%  `;}}
+    html+=bar+'</div><div class="legend">'+legend+'</div>';
+    const ranks=d.step_time.ranks||{};
+    const metrics=['step_time_ms','input_wait_ms','h2d_ms','forward_ms',
+                   'backward_ms','optimizer_ms','ddp_comm_ms','residual_ms'];
+    if(Object.keys(ranks).length){
+      html+='<h2>Step time (ms) by rank</h2><table><tr><th>metric</th>';
+      for(const r of Object.keys(ranks)) html+=`<th>r${r}</th>`;
+      html+='</tr>';
+      for(const m of metrics){
+        let any=false,row=`<tr><td>${m.replace('_ms','')}</td>`;
+        for(const r of Object.keys(ranks)){
+          const v=ranks[r][m];row+=`<td>${v==null?'—':v.toFixed(1)}</td>`;
+          if(v!=null)any=true;}
+        if(any)html+=row+'</tr>';}
+      html+='</table>';
+      html+=`<div class="dim">${d.step_time.steps_analyzed} aligned steps · ${d.step_time.clock} clock · ${d.step_time.strategy}</div>`;
+    }
+    if(d.memory&&Object.keys(d.memory).length){
+      html+='<h2>Peak memory (GiB)</h2><table><tr><th>rank</th><th>allocated</th><th>reserved</th></tr>';
+      for(const [r,m] of Object.entries(d.memory))
+        html+=`<tr><td>r${r}</td><td>${m.alloc==null?'—':(m.alloc/2**30).toFixed(1)}</td><td>${m.reserved==null?'—':(m.reserved/2**30).toFixed(1)}</td></tr>`;
+      html+='</table>';}
+    if(d.system&&d.system.gpus&&Object.keys(d.system.gpus).length){
+      html+='<h2>GPUs (amdsmi)</h2><table><tr><th>gpu</th><th>util %</th><th>VRAM GiB</th><th>temp °C</th><th>power W</th></tr>';
+      for(const [g,m] of Object.entries(d.system.gpus))
+        html+=`<tr><td>${g}</td><td>${m.util==null?'—':m.util.toFixed(0)}</td><td>${m.mem_used==null?'—':(m.mem_used/2**30).toFixed(0)}</td><td>${m.temp==null?'—':m.temp.toFixed(0)}</td><td>${m.power==null?'—':m.power.toFixed(0)}</td></tr>`;
+      html+='</table>';}
+    document.getElementById('content').innerHTML=html;
+  }catch(e){document.getElementById('content').innerHTML='<div class="dim">aggregator not ready…</div>';}
+}
+setInterval(tick,1500);tick();
+</script></body></html>"""
+
+
+def _live_payload(db_path: str) -> dict:
+    from traceml_amd.diagnostics.step_memory.api import load_memory_series
+    from traceml_amd.diagnostics.system.api import load_system_context
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    result = StepTimePipeline(db_path, profile="live").run()
+    window = result.window
+    payload = {
+        "step_time": {
+            "diagnosis": result.diagnosis.primary.to_payload(),
+            "steps_analyzed": window.steps_analyzed,
+            "clock": window.clock,
+            "strategy": window.training_strategy,
+            "shares": window.shares,
+            "ranks": {
+                str(r): window.ranks[r].as_dict() for r in window.ranks_used
+            },
+        },
+        "memory": {},
+        "system": {"gpus": {}},
+    }
+    for rank, series in load_memory_series(db_path).items():
+        alloc = [v for v in series.peak_allocated if v is not None]
+        reserved = [v for v in series.peak_reserved if v is not None]
+        payload["memory"][str(rank)] = {
+            "alloc": max(alloc) if alloc else None,
+            "reserved": max(reserved) if reserved else None,
+        }
+    ctx = load_system_context(db_path)
+    for index, g in ctx.gpus.items():
+        payload["system"]["gpus"][str(index)] = {
+            "util": g.get("util"),
+            "mem_used": g.get("mem_used"),
+            "temp": g.get("temp"),
+            "power": g.get("power"),
+        }
+    return payload
+
+
+class DashboardDisplayDriver(DisplayDriver):
+    def __init__(self, port: int = 8765) -> None:
+        self.port = port
+        self._server = None
+        self._thread: Optional[threading.Thread] = None
+        self._db_path: Optional[str] = None
+
+    def start(self) -> None:
+        try:
+            import uvicorn
+            from fastapi import FastAPI
+            from fastapi.responses import HTMLResponse, JSONResponse
+        except Exception:
+            logger.warning(
+                "traceml_amd: fastapi/uvicorn unavailable, dashboard disabled"
+            )
+            return
+
+        app = FastAPI()
+
+        @app.get("/")
+        def index():
+            return HTMLResponse(_PAGE)
+
+        @app.get("/api/live")
+        def live():
+            if self._db_path is None:
+                return JSONResponse({}, status_code=503)
+            try:
+                return JSONResponse(_live_payload(self._db_path))
+            except Exception as exc:
+                return JSONResponse({"error": repr(exc)}, status_code=500)
+
+        config = uvicorn.Config(
+            app, host="0.0.0.0", port=self.port, log_level="error"
+        )
+        self._server = uvicorn.Server(config)
+        self._thread = threading.Thread(
+            target=self._server.run, name="traceml-dashboard", daemon=True
+        )
+        self._thread.start()
+        print(f"[TraceML-AMD] dashboard at http://localhost:{self.port}", flush=True)
+
+    def render_tick(self, db_path: str) -> None:
+        self._db_path = db_path  # the page polls; nothing to push
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.should_exit = True
+        if self._thread is not None:
+            self._thread.join(timeout=3.0)

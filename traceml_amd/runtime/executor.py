@@ -79,9 +79,11 @@ def main(argv=None) -> int:
     except Exception as exc:  # fail_open=True should prevent this, belt+braces
         _log_crash(settings, "runtime_error", exc)
 
-    # Auto-init instrumentation in run-mode unless the user opted out; a
-    # user's own init() call later is a no-op warn (once-per-process).
-    if os.environ.get("TRACEML_AUTO_INIT", "1") != "0":
+    # The executor starts the RUNTIME only; the instrumentation policy comes
+    # from the user's traceml_amd.init() call (which skips runtime startup
+    # because the handle above is registered). TRACEML_AUTO_INIT=1 opts into
+    # auto-mode init for scripts that never call init() themselves.
+    if os.environ.get("TRACEML_AUTO_INIT", "0") == "1":
         try:
             import traceml_amd
 

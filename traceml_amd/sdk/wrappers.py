@@ -110,7 +110,12 @@ def wrap_backward(loss: Any) -> Any:
 
 def wrap_optimizer(optimizer: Any) -> Any:
     """Wrap optimizer.step() so each call is timed as optimizer_step."""
-    _refuse_if_auto("patch_forward", "optimizer")  # auto mode uses global hooks
+    config = initial.get_active_config()
+    if config is not None and not config.noop and config.mode == "auto":
+        raise RuntimeError(
+            "traceml_amd: optimizer steps are already timed by the global "
+            "optimizer hooks in auto mode; don't wrap the optimizer too"
+        )
     original_step = optimizer.step
 
     def step(*args, **kwargs):
