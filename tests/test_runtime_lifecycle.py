@@ -1,0 +1,206 @@
+"""Lifecycle + budget + durability tests: trace-max-steps DRAINING,
+retention pruning, summary file-RPC, stdout capture, trends, composer."""
+
+import json
+import sqlite3
+import time
+
+import pytest
+import torch
+import torch.nn as nn
+
+
+def test_trace_max_steps_budget(armed_auto_config, tiny_model):
+    from traceml_amd.core import timing
+    from traceml_amd.runtime import state
+    from traceml_amd.sdk.instrumentation import trace_step
+
+    recording = state.recording_state()
+    recording.set_max_steps(3)
+    opt = torch.optim.SGD(tiny_model.parameters(), lr=0.1)
+    for _ in range(6):
+        with trace_step(tiny_model):
+            opt.zero_grad()
+            tiny_model(torch.randn(2, 8)).sum().backward()
+            opt.step()
+    assert recording.phase == state.DRAINING
+    batches = timing.drain_step_time_queue()
+    assert len(batches) == 3  # only the budgeted steps were recorded
+    recording.mark_drained()
+    assert recording.phase == state.COMPLETE
+
+
+def test_retention_pruning(tmp_path):
+    from traceml_amd.aggregator.sqlite_writer import SQLiteWriterSimple
+    import traceml_amd.aggregator.sqlite_writer as sw
+
+    db_path = str(tmp_path / "t.sqlite")
+    writer = SQLiteWriterSimple(db_path)
+    original = sw.RETENTION_ROWS_PER_IDENTITY
+    sw.RETENTION_ROWS_PER_IDENTITY = 50
+    try:
+        writer.start()
+        for i in range(200):
+            writer.ingest(
+                {
+                    "meta": {"sampler": "step_time", "global_rank": 0, "pid": 1},
+                    "body": {
+                        "tables": {
+                            "step_time_samples": [
+                                {"timestamp": time.time(), "step": i, "events": {}}
+                            ]
+                        }
+                    },
+                }
+            )
+        writer.force_flush(timeout=10.0)
+        writer._prune()
+        writer.finalize(budget_sec=10.0)
+        conn = sqlite3.connect(db_path)
+        count = conn.execute("SELECT COUNT(*) FROM step_time_samples").fetchone()[0]
+        newest = conn.execute("SELECT MAX(step) FROM step_time_samples").fetchone()[0]
+        conn.close()
+        assert count == 50  # pruned to retention
+        assert newest == 199  # newest kept
+    finally:
+        sw.RETENTION_ROWS_PER_IDENTITY = original
+
+
+def test_summary_service_file_rpc(tmp_path):
+    from tests import scenarios
+    from traceml_amd.aggregator.summary_service import FinalSummaryService
+    from traceml_amd.sdk import protocol
+    from traceml_amd.utils.atomic_io import atomic_write_json
+
+    session_dir = str(tmp_path)
+    db_path = str(tmp_path / "t.sqlite")
+    scenarios.input_bound(steps=30).write(db_path)
+    service = FinalSummaryService(session_dir, db_path)
+
+    assert service.poll() is False  # no request yet
+    atomic_write_json(
+        protocol.request_path(session_dir), {"request_id": "abc"}
+    )
+    assert service.poll() is True
+    assert (tmp_path / "final_summary.json").exists()
+    response = json.loads(open(protocol.response_path(session_dir)).read())
+    assert response == {"request_id": "abc", "status": "ok"}
+    assert service.poll() is False  # same request id: no rework
+    atomic_write_json(
+        protocol.request_path(session_dir), {"request_id": "def"}
+    )
+    assert service.poll() is True
+
+
+def test_stdout_capture_tee(tmp_path, capsys):
+    from traceml_amd.runtime import stdout_capture
+
+    capture = stdout_capture.install_stream_capture(
+        str(tmp_path / "r0" / "stdout_stderr.log")
+    )
+    print("hello from test")
+    import sys
+
+    print("error line", file=sys.stderr)
+    rows = capture.drain()
+    stdout_capture.reset_for_tests()
+    streams = {(r["stream"], r["line"]) for r in rows}
+    assert ("stdout", "hello from test") in streams
+    assert ("stderr", "error line") in streams
+    log_text = (tmp_path / "r0" / "stdout_stderr.log").read_text()
+    assert "hello from test" in log_text
+    # the user still saw the output
+    captured = capsys.readouterr()
+    assert "hello from test" in captured.out
+
+
+def test_trend_fit_rising_and_flat():
+    from traceml_amd.diagnostics.trends import fit_trend
+
+    rising = fit_trend(list(range(100)), [10.0 + 0.5 * i for i in range(100)])
+    assert rising.direction == "rising"
+    assert rising.delta == pytest.approx(49.5, rel=0.01)
+    flat = fit_trend(list(range(100)), [10.0] * 100)
+    assert flat.direction == "flat"
+    assert fit_trend([1, 2], [1.0, 2.0]) is None  # too few
+
+
+def test_step_time_degrading_issue():
+    from traceml_amd.diagnostics.step_time.trend import step_time_trend_issue
+
+    steps = list(range(1, 61))
+    degrading = [100.0 + 2.0 * s for s in steps]
+    issue = step_time_trend_issue(steps, degrading)
+    assert issue is not None and issue.kind == "STEP_TIME_DEGRADING"
+    stable = [100.0 + (1 if s % 2 else -1) for s in steps]
+    assert step_time_trend_issue(steps, stable) is None
+
+
+def test_model_diagnostics_composer():
+    from traceml_amd.diagnostics.common import DiagnosticIssue, DiagnosticResult
+    from traceml_amd.diagnostics.model_diagnostics import (
+        compose_model_diagnostics,
+        model_card,
+    )
+
+    st = DiagnosticResult(
+        issues=[
+            DiagnosticIssue(kind="INPUT_BOUND", status="INPUT-BOUND",
+                            severity="warn", summary="s", action="a")
+        ]
+    )
+    mem = DiagnosticResult(
+        issues=[
+            DiagnosticIssue(kind="HIGH_MEMORY_PRESSURE",
+                            status="HIGH MEMORY PRESSURE",
+                            severity="crit", summary="m", action="a")
+        ]
+    )
+    combined = compose_model_diagnostics(st, mem)
+    assert combined.primary.kind == "HIGH_MEMORY_PRESSURE"  # crit beats warn
+    assert combined.issues[1].kind == "INPUT_BOUND"
+    assert "HIGH MEMORY PRESSURE" in model_card(combined)
+
+
+def test_cli_view_and_compare(tmp_path, capsys):
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+    from traceml_amd.reporting.final import generate_summary
+
+    db = str(tmp_path / "t.sqlite")
+    scenarios.input_bound(steps=30).write(db)
+    generate_summary(db, str(tmp_path))
+    summary = str(tmp_path / "final_summary.json")
+
+    assert main(["view", summary]) == 0
+    assert "TraceML-AMD Verdict" in capsys.readouterr().out
+
+    assert main(["compare", summary, summary]) == 0
+    out = capsys.readouterr().out
+    assert "Compare Verdict" in out
+
+
+def test_cli_inspect(tmp_path, capsys):
+    from traceml_amd.database.database import Database
+    from traceml_amd.database.writer import DatabaseWriter
+    from traceml_amd.launcher.cli import main
+
+    db = Database()
+    db.add_record("rows", {"a": 1})
+    writer = DatabaseWriter("s", db, str(tmp_path / "data"))
+    writer.flush()
+    writer.close()
+    assert main(["inspect", str(tmp_path)]) == 0
+    out = capsys.readouterr().out
+    assert "rows.msgpack" in out and '"a": 1' in out
+
+
+def test_launch_context_capture(monkeypatch):
+    from traceml_amd.runtime.launch_context import LaunchContext
+
+    monkeypatch.setenv("TRACEML_INTERVAL", "5")
+    monkeypatch.setenv("RANK", "3")
+    ctx = LaunchContext.capture().to_payload()
+    assert ctx["env"]["TRACEML_INTERVAL"] == "5"
+    assert ctx["env"]["RANK"] == "3"
+    assert ctx["cwd"]

@@ -45,6 +45,7 @@ class SQLiteWriterSimple:
         self._thread: Optional[threading.Thread] = None
         self._conn: Optional[sqlite3.Connection] = None
         self._last_prune = 0.0
+        self._conn_lock = threading.Lock()
         self.dropped = 0
         self._prunable_tables: List[str] = []
 
@@ -58,7 +59,9 @@ class SQLiteWriterSimple:
         self._thread.start()
 
     def _open(self) -> None:
-        conn = sqlite3.connect(self.db_path)
+        # finalize()/_prune() may run on the caller's thread after the writer
+        # thread stops; every connection use is serialized by _conn_lock.
+        conn = sqlite3.connect(self.db_path, check_same_thread=False)
         conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA synchronous=NORMAL")
         for writer in self._writers:
@@ -134,17 +137,18 @@ class SQLiteWriterSimple:
                 for sql_table, row in writer.build_rows(item):
                     by_table.setdefault((writer, sql_table), []).append(row)
         if by_table:
-            with conn:
-                for (writer, sql_table), rows in by_table.items():
-                    writer.insert_rows(conn, sql_table, rows)
-                    inserted += len(rows)
+            with self._conn_lock:
+                with conn:
+                    for (writer, sql_table), rows in by_table.items():
+                        writer.insert_rows(conn, sql_table, rows)
+                        inserted += len(rows)
         for barrier in barriers:
             barrier.event.set()
         return inserted
 
     def _prune(self) -> None:
         conn = self._conn
-        with conn:
+        with self._conn_lock, conn:
             for table in self._prunable_tables:
                 conn.execute(
                     f"""
@@ -178,8 +182,9 @@ class SQLiteWriterSimple:
                 if time.time() > deadline:
                     break
             self._prune()
-            self._conn.execute("PRAGMA wal_checkpoint(TRUNCATE)")
-            self._conn.commit()
+            with self._conn_lock:
+                self._conn.execute("PRAGMA wal_checkpoint(TRUNCATE)")
+                self._conn.commit()
         except Exception:
             logger.debug("traceml_amd: sqlite finalize failed", exc_info=True)
         finally:

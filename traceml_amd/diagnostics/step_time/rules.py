@@ -223,8 +223,14 @@ def evaluate(window: StepTimeWindow) -> List[DiagnosticIssue]:
             )
         )
 
-    if not issues:
-        issues.append(
+    # -- degradation trend (supporting issue, never primary on its own) ----
+    trend_issue = _trend_issue(window)
+    if trend_issue is not None:
+        issues.append(trend_issue)
+
+    if not [i for i in issues if i.kind != "STEP_TIME_DEGRADING"]:
+        issues.insert(
+            0,
             DiagnosticIssue(
                 kind="BALANCED",
                 status="BALANCED",
@@ -238,6 +244,21 @@ def evaluate(window: StepTimeWindow) -> List[DiagnosticIssue]:
             )
         )
     return sort_issues(issues)
+
+
+def _trend_issue(window: StepTimeWindow):
+    """Mean-across-ranks per-step step-time series -> degradation check."""
+    from traceml_amd.diagnostics.step_time.trend import step_time_trend_issue
+
+    per_step: dict = {}
+    for series in window.step_series.values():
+        for step, ms in series:
+            per_step.setdefault(step, []).append(ms)
+    if not per_step:
+        return None
+    steps = sorted(per_step)
+    means = [sum(per_step[s]) / len(per_step[s]) for s in steps]
+    return step_time_trend_issue(steps, means)
 
 
 def _straggler_summary(ctx: StragglerContext) -> str:

@@ -63,6 +63,8 @@ def launch_process(
     env = dict(os.environ)
     apply_settings_to_env(settings, env)
 
+    from traceml_amd.runtime.launch_context import LaunchContext
+
     manifest_mod.write_run_manifest(
         sdir,
         manifest_mod.STATUS_STARTING,
@@ -71,6 +73,7 @@ def launch_process(
         world_size=nnodes * nproc_per_node,
         nnodes=nnodes,
         run_name=settings.run_name,
+        extra={"launch_context": LaunchContext.capture().to_payload()},
     )
     manifest_mod.write_code_manifest(sdir, script)
 

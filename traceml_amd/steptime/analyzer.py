@@ -86,7 +86,9 @@ class StepTimeAnalyzer:
         total_cells = len(used_ranks) * len(steps)
         for rank in used_ranks:
             rank_rows = [by_rank[rank][s] for s in steps]
-            window.ranks[rank] = self._rank_values(rank_rows, clock)
+            values, series = self._rank_values(rank_rows, clock)
+            window.ranks[rank] = values
+            window.step_series[rank] = series
             window.identities[rank] = self._identity(rank_rows[-1])
             for signal in _CORE_SIGNALS:
                 coverage_hits[signal] += sum(
@@ -173,7 +175,7 @@ class StepTimeAnalyzer:
 
     def _rank_values(
         self, rank_rows: List[StepTimeSourceRow], clock: str
-    ) -> StepTimeValues:
+    ) -> tuple:
         n = len(rank_rows)
 
         def cell(row, signal) -> Optional[dict]:
@@ -251,7 +253,12 @@ class StepTimeAnalyzer:
             # every-step semantics: a metric is the window mean of its
             # measured steps; never fabricated from missing cells
             means[metric] = (sum(samples) / len(samples)) if samples else None
-        return StepTimeValues(**means)
+        series = [
+            (row.step, values["step_time_ms"])
+            for row, values in zip(rank_rows, per_step)
+            if values.get("step_time_ms") is not None
+        ]
+        return StepTimeValues(**means), series
 
     @staticmethod
     def _aggregate(window: StepTimeWindow) -> None:

@@ -136,6 +136,8 @@ class StepTimeWindow:
     worst: Dict[str, Optional[dict]] = field(default_factory=dict)
     #: phase -> share of average step_time_ms (observational)
     shares: Dict[str, Optional[float]] = field(default_factory=dict)
+    #: rank -> [(step, step_time_ms), ...] for trend detection
+    step_series: Dict[int, List] = field(default_factory=dict)
 
     @property
     def ranks_used(self) -> List[int]:
