@@ -198,3 +198,32 @@ def test_mark_overhead_under_10us(ext):
     host_us = (time.perf_counter() - t0) * 1e6 / n
     torch.cuda.synchronize()
     assert host_us < 25.0, f"ring_mark host cost {host_us:.1f}us"
+
+
+@requires_gpu
+def test_cross_stream_stamps_consistent(ext):
+    """The ddp_comm hook pattern: start stamp on the main stream, end stamp
+    on a side stream that waits on the work — s_memrealtime is globally
+    consistent across streams so the pair must bracket the kernel time."""
+    main = torch.cuda.current_stream()
+    side = torch.cuda.Stream()
+    x = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+    torch.cuda.synchronize()
+
+    e0 = ext.event_acquire()
+    e1 = ext.event_acquire()
+    s0 = ext.ring_mark(main.cuda_stream)
+    ext.event_record(e0, main.cuda_stream)
+    for _ in range(10):
+        y = x @ x
+    ext.event_record(e1, main.cuda_stream)
+    side.wait_stream(main)  # side stream waits for the GEMMs
+    s1 = ext.ring_mark(side.cuda_stream)
+    torch.cuda.synchronize()
+
+    ring_ms = ext.ring_elapsed_ms(s0, s1)
+    event_ms = ext.event_elapsed_ms(e0, e1)
+    ext.event_release(e0)
+    ext.event_release(e1)
+    assert ring_ms >= event_ms * 0.9  # must cover the kernels
+    assert ring_ms < event_ms + 5.0, (ring_ms, event_ms)  # and stay tight
