@@ -227,3 +227,43 @@ def test_cross_stream_stamps_consistent(ext):
     ext.event_release(e1)
     assert ring_ms >= event_ms * 0.9  # must cover the kernels
     assert ring_ms < event_ms + 5.0, (ring_ms, event_ms)  # and stay tight
+
+
+@requires_gpu
+def test_fsdp_wrap_traced_on_gpu(armed_auto_config):
+    """Real FSDP wrap (needs an accelerator): forward timed once per step,
+    strategy detected as fsdp (world_size=1 process group on RCCL)."""
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        import os
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29630")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
+
+    from tests.conftest import drain_step_time_rows
+    from traceml_amd.core import event_names
+    from traceml_amd.runtime.environment import detect_runtime_environment
+    from traceml_amd.runtime.identity import RuntimeIdentity
+    from traceml_amd.sdk.instrumentation import trace_step
+
+    inner = torch.nn.Sequential(
+        torch.nn.Linear(64, 128), torch.nn.ReLU(), torch.nn.Linear(128, 8)
+    ).cuda()
+    model = FSDP(inner)
+    info = detect_runtime_environment(RuntimeIdentity(world_size=1), model)
+    assert info.training_strategy == "fsdp"
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    for _ in range(3):
+        with trace_step(model):
+            opt.zero_grad()
+            model(torch.randn(16, 64, device="cuda")).sum().backward()
+            opt.step()
+    torch.cuda.synchronize()
+    rows = drain_step_time_rows()
+    assert len(rows) == 3
+    for row in rows:
+        assert row["events"][event_names.FORWARD]["n_calls"] == 1
+    dist.destroy_process_group()
