@@ -1,0 +1,74 @@
+"""Renderer view models + live-session freshness states."""
+
+import pytest
+
+from tests import scenarios
+from traceml_amd.renderers import live_view
+from traceml_amd.steptime.pipeline import (
+    FRESH_BRIDGED,
+    FRESH_COLD,
+    FRESH_EXPIRED,
+    FRESH_LIVE,
+    LiveStepTimeSession,
+)
+
+
+@pytest.fixture
+def db_path(tmp_path):
+    return str(tmp_path / "t.sqlite")
+
+
+def test_live_view_payload_shape(db_path):
+    scenarios.input_straggler(steps=30).write(db_path)
+    payload = live_view(db_path)
+    assert payload["freshness"] == "live"
+    st = payload["step_time"]
+    assert st["diagnosis"]["kind"] == "INPUT_STRAGGLER"
+    assert set(st["ranks"]) == {"0", "1", "2", "3"}
+    assert st["ranks"]["2"]["input_wait_ms"] > 100
+    assert "memory" in payload and "system" in payload
+
+
+def test_live_session_freshness_transitions(db_path, tmp_path):
+    scenarios.StepTimeScenario("empty", {}, steps=0).write(db_path)
+    session = LiveStepTimeSession(db_path, ttl_sec=0.2)
+    _, freshness = session.tick()
+    assert freshness == FRESH_COLD  # no data ever
+
+    scenarios.healthy_ddp(ranks=1, steps=10).write(db_path)
+    result, freshness = session.tick()
+    assert freshness == FRESH_LIVE
+    assert result.window.steps_analyzed == 10
+
+    # wipe the table -> transient empty read is bridged by the last good
+    import sqlite3
+
+    conn = sqlite3.connect(db_path)
+    conn.execute("DELETE FROM step_time_samples")
+    conn.commit()
+    conn.close()
+    result, freshness = session.tick()
+    assert freshness == FRESH_BRIDGED
+    assert result.window.steps_analyzed == 10  # last good served
+
+    import time
+
+    time.sleep(0.25)
+    _, freshness = session.tick()
+    assert freshness == FRESH_EXPIRED
+
+
+def test_cli_driver_renders_without_terminal(db_path):
+    scenarios.healthy_ddp(ranks=2, steps=30).write(db_path)
+    from traceml_amd.aggregator.display.cli import CLIDisplayDriver
+
+    driver = CLIDisplayDriver()
+    panel = driver._build(db_path)  # builds the Rich renderable directly
+    from rich.console import Console
+    import io
+
+    console = Console(file=io.StringIO(), width=120)
+    console.print(panel)
+    text = console.file.getvalue()
+    assert "traceml-amd live" in text
+    assert "backward" in text

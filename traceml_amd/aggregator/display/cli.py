@@ -20,6 +20,7 @@ class CLIDisplayDriver(DisplayDriver):
     def __init__(self) -> None:
         self._live = None
         self._console = None
+        self._session = None
 
     def start(self) -> None:
         try:
@@ -50,9 +51,11 @@ class CLIDisplayDriver(DisplayDriver):
         from rich.table import Table
         from rich.console import Group
 
-        from traceml_amd.steptime.pipeline import StepTimePipeline
+        from traceml_amd.steptime.pipeline import LiveStepTimeSession
 
-        result = StepTimePipeline(db_path, profile="live").run()
+        if self._session is None:
+            self._session = LiveStepTimeSession(db_path)
+        result, freshness = self._session.tick()
         window = result.window
         diag = result.diagnosis.primary
 
@@ -60,6 +63,8 @@ class CLIDisplayDriver(DisplayDriver):
             f"[{_SEVERITY_STYLE.get(diag.severity, 'cyan')}]"
             f"{diag.status}[/]  {diag.summary}"
         )
+        if freshness not in ("live", "cold"):
+            header += f"  [dim]({freshness})[/dim]"
         renderables = [header]
 
         if window.has_data:

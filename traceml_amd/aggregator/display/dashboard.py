@@ -84,43 +84,10 @@ setInterval(tick,1500);tick();
 </script></body></html>"""
 
 
-def _live_payload(db_path: str) -> dict:
-    from traceml_amd.diagnostics.step_memory.api import load_memory_series
-    from traceml_amd.diagnostics.system.api import load_system_context
-    from traceml_amd.steptime.pipeline import StepTimePipeline
+def _live_payload(db_path: str, session=None) -> dict:
+    from traceml_amd.renderers import live_view
 
-    result = StepTimePipeline(db_path, profile="live").run()
-    window = result.window
-    payload = {
-        "step_time": {
-            "diagnosis": result.diagnosis.primary.to_payload(),
-            "steps_analyzed": window.steps_analyzed,
-            "clock": window.clock,
-            "strategy": window.training_strategy,
-            "shares": window.shares,
-            "ranks": {
-                str(r): window.ranks[r].as_dict() for r in window.ranks_used
-            },
-        },
-        "memory": {},
-        "system": {"gpus": {}},
-    }
-    for rank, series in load_memory_series(db_path).items():
-        alloc = [v for v in series.peak_allocated if v is not None]
-        reserved = [v for v in series.peak_reserved if v is not None]
-        payload["memory"][str(rank)] = {
-            "alloc": max(alloc) if alloc else None,
-            "reserved": max(reserved) if reserved else None,
-        }
-    ctx = load_system_context(db_path)
-    for index, g in ctx.gpus.items():
-        payload["system"]["gpus"][str(index)] = {
-            "util": g.get("util"),
-            "mem_used": g.get("mem_used"),
-            "temp": g.get("temp"),
-            "power": g.get("power"),
-        }
-    return payload
+    return live_view(db_path, session=session)
 
 
 class DashboardDisplayDriver(DisplayDriver):
@@ -129,6 +96,7 @@ class DashboardDisplayDriver(DisplayDriver):
         self._server = None
         self._thread: Optional[threading.Thread] = None
         self._db_path: Optional[str] = None
+        self._session = None
 
     def start(self) -> None:
         try:
@@ -152,7 +120,11 @@ class DashboardDisplayDriver(DisplayDriver):
             if self._db_path is None:
                 return JSONResponse({}, status_code=503)
             try:
-                return JSONResponse(_live_payload(self._db_path))
+                if self._session is None:
+                    from traceml_amd.steptime.pipeline import LiveStepTimeSession
+
+                    self._session = LiveStepTimeSession(self._db_path)
+                return JSONResponse(_live_payload(self._db_path, self._session))
             except Exception as exc:
                 return JSONResponse({"error": repr(exc)}, status_code=500)
 
