@@ -207,6 +207,7 @@ def main():
     os.environ["TRACEML_LOGS_DIR"] = logs_dir
     os.environ["TRACEML_SESSION_ID"] = session_id
     os.environ["TRACEML_INTERVAL"] = "1.0"
+    os.environ["TRACEML_FINALIZE_TIMEOUT"] = "30"  # bench must exit promptly
     os.environ["TRACEML_AGGREGATOR_PORT"] = os.environ.get(
         "TRACEML_AGGREGATOR_PORT", "29877"
     )
