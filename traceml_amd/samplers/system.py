@@ -112,15 +112,23 @@ class _AmdSmi:
             pass
         try:
             power = smi.amdsmi_get_power_info(h)
-            watts = power.get("average_socket_power") or power.get(
-                "current_socket_power"
+
+            def _num(value):
+                # amdsmi returns the literal string 'N/A' for absent fields
+                if value is None or isinstance(value, str):
+                    return None
+                return float(value)
+
+            watts = (
+                _num(power.get("average_socket_power"))
+                or _num(power.get("current_socket_power"))
+                or _num(power.get("socket_power"))
             )
-            if watts is not None and watts != "N/A":
-                row["power_w"] = float(watts)
-            cap = power.get("power_limit")
-            if cap is not None and cap != "N/A":
-                # amdsmi reports power_limit in W (some versions in uW)
-                cap = float(cap)
+            if watts is not None:
+                row["power_w"] = watts
+            cap = _num(power.get("power_limit"))
+            if cap is not None:
+                # MI355X reports power_limit in uW (1400000000 -> 1400 W)
                 row["power_cap_w"] = cap / 1e6 if cap > 1e5 else cap
         except Exception:
             pass
