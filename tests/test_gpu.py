@@ -208,6 +208,11 @@ def test_cross_stream_stamps_consistent(ext):
     main = torch.cuda.current_stream()
     side = torch.cuda.Stream()
     x = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+    # warm both streams (side-stream first use costs ms-scale setup)
+    for _ in range(3):
+        x @ x
+        ext.ring_mark(main.cuda_stream)
+        ext.ring_mark(side.cuda_stream)
     torch.cuda.synchronize()
 
     e0 = ext.event_acquire()
@@ -221,12 +226,13 @@ def test_cross_stream_stamps_consistent(ext):
     s1 = ext.ring_mark(side.cuda_stream)
     torch.cuda.synchronize()
 
+    assert ext.ring_ready(s0) and ext.ring_ready(s1), "stamps did not land"
     ring_ms = ext.ring_elapsed_ms(s0, s1)
     event_ms = ext.event_elapsed_ms(e0, e1)
     ext.event_release(e0)
     ext.event_release(e1)
-    assert ring_ms >= event_ms * 0.9  # must cover the kernels
-    assert ring_ms < event_ms + 5.0, (ring_ms, event_ms)  # and stay tight
+    assert ring_ms >= event_ms * 0.9, (ring_ms, event_ms)  # covers the kernels
+    assert ring_ms < event_ms + 5.0, (ring_ms, event_ms)  # and stays tight
 
 
 @requires_gpu
