@@ -6,7 +6,10 @@ SwiGLU, RoPE, GQA)."""
 from __future__ import annotations
 
 
-def build_llama3(tiny: bool = False, seq_len: int = 8192):
+def build_llama3(tiny: bool = False, seq_len: int = 8192, device=None):
+    """device="cuda" initializes weights directly on the GPU (an 8B random
+    init on CPU costs minutes; on-device it is seconds)."""
+    import torch
     from transformers import LlamaConfig, LlamaForCausalLM
 
     if tiny:
@@ -32,4 +35,7 @@ def build_llama3(tiny: bool = False, seq_len: int = 8192):
             rope_theta=500000.0,
         )
     config.use_cache = False
+    if device is not None:
+        with torch.device(device):
+            return LlamaForCausalLM(config)
     return LlamaForCausalLM(config)
