@@ -155,3 +155,26 @@ def test_rank_memory_imbalance(db_path):
     result = diagnose_step_memory(load_memory_series(db_path))
     kinds = [i.kind for i in result.issues]
     assert "RANK_MEMORY_IMBALANCE" in kinds
+
+
+def test_eight_rank_healthy_balanced(db_path):
+    """Full-node shape (8 GPUs, one process per GPU over xGMI)."""
+    scenarios.healthy_ddp(ranks=8, steps=30, gpu=True).write(db_path)
+    result = run_pipeline(db_path)
+    assert result.diagnosis.primary.kind in ("BALANCED", "COMPUTE_BOUND")
+    assert result.window.ranks_used == list(range(8))
+    assert result.window.clock == "gpu"
+
+
+def test_eight_rank_straggler_any_position(db_path):
+    """Culprit detection must not depend on the straggler's rank index."""
+    profiles = {}
+    for r in range(8):
+        if r == 6:
+            profiles[r] = scenarios.RankProfile(input_ms=184.0, backward_ms=55.0)
+        else:
+            profiles[r] = scenarios.RankProfile(input_ms=4.0, backward_ms=235.0)
+    scenarios.StepTimeScenario("s8", profiles, steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "INPUT_STRAGGLER"
+    assert primary.ranks == [6]

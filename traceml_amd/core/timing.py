@@ -32,7 +32,7 @@ STEP_TIME_QUEUE_MAX = 2048
 _CPU_ONLY_EVENTS = frozenset({event_names.DATALOADER})
 
 
-@dataclass
+@dataclass(slots=True)
 class TimeEvent:
     name: str
     device: str
@@ -182,13 +182,20 @@ def flush_step_time_buffer(step: int) -> None:
             return
         events = list(_step_buffer)
         _step_buffer.clear()
-    summary: dict = {}
-    for event in events:
-        key = _SUMMARY_KEYS.get(event.name)
-        if key is not None and event.cpu_ms is not None:
-            summary[key] = summary.get(key, 0.0) + event.cpu_ms
-    _last_cpu_summary.clear()
-    _last_cpu_summary.update(summary)
+    # the CPU summary feeds the RCCL rank-stats gather; skip the work
+    # entirely when no exchange is active (single-process runs)
+    from traceml_amd.parallel.rank_stats import get_active_exchange
+
+    if get_active_exchange() is not None:
+        summary: dict = {}
+        for event in events:
+            key = _SUMMARY_KEYS.get(event.name)
+            if key is not None and event.cpu_end is not None:
+                summary[key] = summary.get(key, 0.0) + (
+                    (event.cpu_end - event.cpu_start) * 1000.0
+                )
+        _last_cpu_summary.clear()
+        _last_cpu_summary.update(summary)
     batch = StepTimeBatch(step=step, events=events, flushed_at=time.time())
     with _queue_lock:
         if len(_step_time_queue) >= STEP_TIME_QUEUE_MAX:
