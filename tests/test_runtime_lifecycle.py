@@ -204,3 +204,26 @@ def test_launch_context_capture(monkeypatch):
     assert ctx["env"]["TRACEML_INTERVAL"] == "5"
     assert ctx["env"]["RANK"] == "3"
     assert ctx["cwd"]
+
+
+def test_export_chrome_trace(tmp_path, capsys):
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+
+    db = str(tmp_path / "t.sqlite")
+    scenarios.input_straggler(steps=10).write(db)
+    out = str(tmp_path / "trace.json")
+    assert main(["export-trace", db, "-o", out]) == 0
+    payload = json.loads(open(out).read())
+    events = payload["traceEvents"]
+    phase_events = [e for e in events if e.get("ph") == "X"]
+    # 4 ranks x 10 steps, each with a step envelope + phases (+ ddp_comm)
+    assert len(phase_events) > 4 * 10 * 4
+    assert {e["pid"] for e in phase_events} == {0, 1, 2, 3}
+    assert any(e["name"] == "ddp_comm" for e in phase_events)
+    # rank 2's input_wait dwarfs the others (the straggler is visible)
+    r2_input = [e["dur"] for e in phase_events
+                if e["pid"] == 2 and e["name"] == "input_wait"]
+    r0_input = [e["dur"] for e in phase_events
+                if e["pid"] == 0 and e["name"] == "input_wait"]
+    assert min(r2_input) > 10 * max(r0_input)

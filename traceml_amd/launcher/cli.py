@@ -85,6 +85,14 @@ def build_parser() -> argparse.ArgumentParser:
     inspect_p = sub.add_parser("inspect", help="dump per-rank msgpack backups")
     inspect_p.add_argument("path")
 
+    trace_p = sub.add_parser(
+        "export-trace",
+        help="export the step-time history as a chrome://tracing timeline",
+    )
+    trace_p.add_argument("telemetry_sqlite")
+    trace_p.add_argument("-o", "--output", default="trace.json")
+    trace_p.add_argument("--max-steps", type=int, default=None)
+
     return parser
 
 
@@ -117,6 +125,14 @@ def main(argv: Optional[List[str]] = None) -> int:
         return commands.run_view(args.summary_json)
     if args.command == "inspect":
         return commands.run_inspect(args.path)
+    if args.command == "export-trace":
+        from traceml_amd.reporting.trace_export import export_chrome_trace
+
+        n = export_chrome_trace(
+            args.telemetry_sqlite, args.output, max_steps=args.max_steps
+        )
+        print(f"wrote {n} trace events to {args.output}")
+        return 0
     return 2
 
 
