@@ -177,15 +177,15 @@ class StepTimeAnalyzer:
         self, rank_rows: List[StepTimeSourceRow], clock: str
     ) -> tuple:
         n = len(rank_rows)
-
-        def cell(row, signal) -> Optional[dict]:
-            return row.events.get(signal)
+        use_gpu_clock = clock == "gpu"
 
         def sel_ms(c: Optional[dict]) -> Optional[float]:
             if c is None:
                 return None
-            if clock == "gpu" and c.get("gpu_ms") is not None:
-                return float(c["gpu_ms"])
+            if use_gpu_clock:
+                gpu = c.get("gpu_ms")
+                if gpu is not None:
+                    return float(gpu)
             return _f(c.get("cpu_ms"))
 
         def cpu_ms(c: Optional[dict]) -> Optional[float]:
@@ -200,22 +200,30 @@ class StepTimeAnalyzer:
             available[signal] = all(signal in r.events for r in rank_rows)
         for signal in OCCURRENCE_SIGNALS:
             available[signal] = any(signal in r.events for r in rank_rows)
+        has_input = available["dataloader"]
+        has_traced = available["traced"]
+        has_h2d = available["h2d"]
+        has_fwd = available["forward"]
+        has_bwd = available["backward"]
+        has_opt = available["optimizer"]
+        has_ddp = available["ddp_comm"]
 
         per_step: List[dict] = []
         for row in rank_rows:
+            events = row.events
             values: dict = {}
-            input_cpu = cpu_ms(cell(row, "dataloader")) if available["dataloader"] else None
-            traced_c = cell(row, "traced") if available["traced"] else None
+            input_cpu = cpu_ms(events.get("dataloader")) if has_input else None
+            traced_c = events.get("traced") if has_traced else None
             traced_sel = sel_ms(traced_c)
             traced_cpu = cpu_ms(traced_c)
             traced_gpu = gpu_ms(traced_c)
-            h2d = sel_ms(cell(row, "h2d")) if available["h2d"] else None
+            h2d = sel_ms(events.get("h2d")) if has_h2d else None
             h2d0 = h2d if h2d is not None else 0.0
-            fwd = sel_ms(cell(row, "forward")) if available["forward"] else None
-            bwd = sel_ms(cell(row, "backward")) if available["backward"] else None
-            opt = sel_ms(cell(row, "optimizer")) if available["optimizer"] else None
+            fwd = sel_ms(events.get("forward")) if has_fwd else None
+            bwd = sel_ms(events.get("backward")) if has_bwd else None
+            opt = sel_ms(events.get("optimizer")) if has_opt else None
             opt0 = opt if opt is not None else 0.0
-            ddp = sel_ms(cell(row, "ddp_comm")) if available["ddp_comm"] else None
+            ddp = sel_ms(events.get("ddp_comm")) if has_ddp else None
 
             compute = None
             if fwd is not None and bwd is not None:

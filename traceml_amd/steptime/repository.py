@@ -23,25 +23,40 @@ def _connect(db_path: str) -> sqlite3.Connection:
 
 
 def normalize_step_time_events(raw: Optional[str]) -> Optional[dict]:
-    """events_json -> {signal_key: {duration_ms, cpu_ms, gpu_ms, n_calls, is_gpu}}."""
+    """events_json -> {signal_key: {duration_ms, cpu_ms, gpu_ms, n_calls, is_gpu}}.
+
+    Hot path for summary loads (10k+ rows): json values are already
+    numbers/None from our own writer, so coercion only runs for the
+    defensive string/garbage case.
+    """
     if not raw:
         return None
     try:
         decoded = json.loads(raw)
     except (ValueError, TypeError):
         return None
-    if not isinstance(decoded, dict):
+    if type(decoded) is not dict:
         return None
     events = {}
+    names = STEP_TIME_EVENT_NAMES
     for wire_name, cell in decoded.items():
-        signal = STEP_TIME_EVENT_NAMES.get(wire_name)
-        if signal is None or not isinstance(cell, dict):
+        signal = names.get(wire_name)
+        if signal is None or type(cell) is not dict:
             continue
+        duration = cell.get("duration_ms")
+        cpu = cell.get("cpu_ms")
+        gpu = cell.get("gpu_ms")
+        if isinstance(duration, str):
+            duration = _num(duration)
+        if isinstance(cpu, str):
+            cpu = _num(cpu)
+        if isinstance(gpu, str):
+            gpu = _num(gpu)
         events[signal] = {
-            "duration_ms": _num(cell.get("duration_ms")),
-            "cpu_ms": _num(cell.get("cpu_ms")),
-            "gpu_ms": _num(cell.get("gpu_ms")),
-            "n_calls": int(cell.get("n_calls") or 0),
+            "duration_ms": duration,
+            "cpu_ms": cpu,
+            "gpu_ms": gpu,
+            "n_calls": cell.get("n_calls") or 0,
             "is_gpu": bool(cell.get("is_gpu")),
         }
     return events
