@@ -178,3 +178,73 @@ def test_eight_rank_straggler_any_position(db_path):
     primary = run_pipeline(db_path).diagnosis.primary
     assert primary.kind == "INPUT_STRAGGLER"
     assert primary.ranks == [6]
+
+
+def test_h2d_bound(db_path):
+    profiles = {0: scenarios.RankProfile(input_ms=2.0, h2d_ms=40.0,
+                                         forward_ms=30.0, backward_ms=55.0,
+                                         optimizer_ms=8.0)}
+    scenarios.StepTimeScenario("h2d", profiles, steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "H2D_BOUND"
+    assert primary.severity == "crit"  # ~30% of step time
+
+
+def test_h2d_straggler(db_path):
+    profiles = {}
+    for r in range(4):
+        if r == 3:
+            profiles[r] = scenarios.RankProfile(h2d_ms=180.0, backward_ms=55.0)
+        else:
+            profiles[r] = scenarios.RankProfile(h2d_ms=0.5, backward_ms=235.0)
+    scenarios.StepTimeScenario("h2ds", profiles, steps=30).write(db_path)
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "H2D_STRAGGLER"
+    assert primary.ranks == [3]
+
+
+def test_incomplete_data_missing_forward(db_path):
+    """Forward patch disabled -> forward never measured -> INCOMPLETE_DATA
+    with the missing signal named (the incomplete_signals demo contract)."""
+    import json as _json
+    import sqlite3 as _sqlite3
+    import time as _time
+
+    from traceml_amd.aggregator.writers import build_all_writers
+    from traceml_amd.core import event_names
+
+    conn = _sqlite3.connect(db_path)
+    for w in build_all_writers():
+        w.init_schema(conn)
+    profile = scenarios.RankProfile()
+    with conn:
+        for step in range(1, 31):
+            events = profile.events()
+            del events[event_names.FORWARD]
+            conn.execute(
+                "INSERT INTO step_time_samples (global_rank, world_size, "
+                "timestamp, step, events_json) VALUES (0, 1, ?, ?, ?)",
+                (_time.time() + step * 0.1, step, _json.dumps(events)),
+            )
+    conn.close()
+    primary = run_pipeline(db_path).diagnosis.primary
+    assert primary.kind == "INCOMPLETE_DATA"
+    assert "forward" in primary.evidence["missing_signals"]
+    assert primary.evidence["signal_coverage"]["forward"] == 0.0
+
+
+def test_straggler_outranks_share_issue(db_path):
+    """When a straggler and a phase-share issue coexist, the more severe /
+    higher-scoring wins issues[0] (severity-then-score ordering)."""
+    profiles = {}
+    for r in range(4):
+        if r == 1:
+            profiles[r] = scenarios.RankProfile(input_ms=300.0, backward_ms=55.0)
+        else:
+            # every rank also has an elevated input share (~14% of step)
+            profiles[r] = scenarios.RankProfile(input_ms=60.0, backward_ms=350.0)
+    scenarios.StepTimeScenario("mix", profiles, steps=30).write(db_path)
+    result = run_pipeline(db_path)
+    kinds = [i.kind for i in result.diagnosis.issues]
+    assert result.diagnosis.primary.kind == "INPUT_STRAGGLER"
+    assert "INPUT_BOUND" in kinds  # the shared input-share issue still listed
