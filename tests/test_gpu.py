@@ -273,3 +273,62 @@ def test_fsdp_wrap_traced_on_gpu(armed_auto_config):
     for row in rows:
         assert row["events"][event_names.FORWARD]["n_calls"] == 1
     dist.destroy_process_group()
+
+
+@requires_gpu
+def test_ring_wrap_small_ring_subprocess():
+    """Force ring wraps (1024 slots, ~70 marks/wrap-window) in a traced
+    500-step loop: every step must still resolve (the sampler keeps up) and
+    no batch may wedge the queue."""
+    import os
+    import subprocess
+    import sys
+    import textwrap
+
+    script = textwrap.dedent(
+        """
+        import sys, os
+        sys.path.insert(0, os.environ["REPO"])
+        import torch
+        from traceml_amd.runtime.settings import TraceMLSettings
+        from traceml_amd.sdk import initial
+        from traceml_amd.sdk.instrumentation import trace_step
+        from traceml_amd.samplers.step_time import StepTimeSampler
+        from traceml_amd.database.database import Database
+        from traceml_amd.models.mlp import TinyMLP
+
+        cfg = initial._build_config("auto", None, None, None, None, TraceMLSettings())
+        initial._apply_requested_patches(cfg); initial._active_config = cfg
+        model = TinyMLP().cuda()
+        opt = torch.optim.SGD(model.parameters(), lr=0.01)
+        db = Database(maxlen=10000)
+        sampler = StepTimeSampler(db)
+        for i in range(500):
+            with trace_step(model):
+                x = torch.randn(64, 256).cuda()
+                opt.zero_grad()
+                model(x).sum().backward()
+                opt.step()
+            if i % 10 == 0:
+                sampler.sample()
+        torch.cuda.synchronize()
+        sampler.sample()
+        rows = db.tail("step_time_samples")
+        assert len(rows) == 500, f"resolved {len(rows)} of 500 steps"
+        gpu_rows = sum(
+            1 for r in rows
+            if r["events"]["_traceml_internal:forward_time"]["gpu_ms"] is not None
+        )
+        assert gpu_rows >= 490, f"only {gpu_rows} rows kept the GPU clock"
+        print("WRAP_OK", len(rows), gpu_rows)
+        """
+    )
+    env = dict(os.environ)
+    env["REPO"] = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["TRACEML_AMD_RING_SLOTS"] = "1024"
+    proc = subprocess.run(
+        [sys.executable, "-c", script], env=env, capture_output=True,
+        text=True, timeout=300,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "WRAP_OK" in proc.stdout
