@@ -85,3 +85,31 @@ def test_final_summary_smoke(tmp_path):
     rank_data = session / "r0" / "data"
     assert rank_data.is_dir()
     assert any(rank_data.rglob("*.msgpack"))
+
+
+@pytest.mark.timeout(240)
+def test_crash_stderr_tail_written(tmp_path):
+    script = tmp_path / "crash.py"
+    script.write_text("raise RuntimeError('deliberate crash for tail test')\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "20"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "crash",
+            "--aggregator-port", "29878",
+            "--master-port", "29602",
+            str(script),
+        ],
+        env=env, capture_output=True, text=True, timeout=220, cwd=REPO_ROOT,
+    )
+    assert proc.returncode != 0
+    session = tmp_path / "logs" / "crash"
+    tail_log = session / "crash_stderr.log"
+    assert tail_log.exists()
+    assert "deliberate crash for tail test" in tail_log.read_text()
+    manifest = json.loads((session / "manifest.json").read_text())
+    assert manifest["status"] == "failed"

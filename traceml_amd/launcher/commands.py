@@ -108,8 +108,8 @@ def launch_process(
             script_args=script_args,
         )
         manifest_mod.update_status(sdir, manifest_mod.STATUS_RUNNING)
-        train_proc, _ = spawn_process_group(
-            torchrun.to_command(_executor_path()), env=env
+        train_proc, stderr_tail = spawn_process_group(
+            torchrun.to_command(_executor_path()), env=env, capture_stderr=True
         )
 
         # Monitor loop: if the aggregator dies early, training continues
@@ -123,6 +123,16 @@ def launch_process(
                 agg_proc = None
             time.sleep(0.5)
         train_code = train_proc.returncode
+        if train_code != 0 and stderr_tail is not None:
+            # crash forensics: last 64 KiB of the training tree's stderr
+            stderr_tail.join(timeout=2.0)
+            try:
+                with open(
+                    os.path.join(sdir, "crash_stderr.log"), "w", encoding="utf-8"
+                ) as f:
+                    f.write(stderr_tail.tail())
+            except OSError:
+                pass
     finally:
         if agg_proc is not None:
             terminate_process_group(
