@@ -110,6 +110,10 @@ def compare_payloads(baseline: dict, candidate: dict) -> dict:
         "verdict": verdict,
         "baseline_diagnosis": baseline.get("primary_diagnosis", {}).get("kind"),
         "candidate_diagnosis": candidate.get("primary_diagnosis", {}).get("kind"),
+        "diagnosis_transition": _diagnosis_transition(baseline, candidate),
+        "per_rank_step_time": _per_rank_compare(
+            baseline, candidate, "step_time", "step_time_ms"
+        ),
         "metrics": [
             {
                 "section": d.section,
@@ -126,11 +130,65 @@ def compare_payloads(baseline: dict, candidate: dict) -> dict:
     }
 
 
+_SEVERITY_RANK = {"info": 0, "warn": 1, "crit": 2}
+
+
+def _diagnosis_transition(baseline: dict, candidate: dict) -> dict:
+    b = baseline.get("primary_diagnosis", {})
+    c = candidate.get("primary_diagnosis", {})
+    b_sev = _SEVERITY_RANK.get(b.get("severity"), 0)
+    c_sev = _SEVERITY_RANK.get(c.get("severity"), 0)
+    if c_sev < b_sev:
+        direction = "resolved" if c_sev == 0 else "improved"
+    elif c_sev > b_sev:
+        direction = "worsened"
+    elif b.get("kind") != c.get("kind"):
+        direction = "changed"
+    else:
+        direction = "unchanged"
+    return {
+        "from": {"kind": b.get("kind"), "severity": b.get("severity")},
+        "to": {"kind": c.get("kind"), "severity": c.get("severity")},
+        "direction": direction,
+    }
+
+
+def _per_rank_compare(
+    baseline: dict, candidate: dict, section: str, metric: str
+) -> List[dict]:
+    def rank_values(payload):
+        rows = payload.get(section, {}).get("groups", {}).get("rows", {})
+        return {
+            key: row.get("metrics", {}).get(metric)
+            for key, row in rows.items()
+        }
+
+    b_rows = rank_values(baseline)
+    c_rows = rank_values(candidate)
+    out = []
+    for key in sorted(set(b_rows) | set(c_rows), key=lambda k: (len(k), k)):
+        b = b_rows.get(key)
+        c = c_rows.get(key)
+        delta = (c - b) if (b is not None and c is not None) else None
+        out.append(
+            {
+                "rank": key,
+                "baseline": b,
+                "candidate": c,
+                "delta": delta,
+                "pct": (delta / b * 100.0) if (delta is not None and b) else None,
+            }
+        )
+    return out
+
+
 def render_compare(result: dict) -> str:
+    transition = result.get("diagnosis_transition", {})
     lines = [
         f"TraceML-AMD Compare Verdict: {result['verdict']}",
         f"  diagnosis: {result['baseline_diagnosis']} -> "
-        f"{result['candidate_diagnosis']}",
+        f"{result['candidate_diagnosis']}"
+        + (f"  ({transition.get('direction')})" if transition else ""),
         "",
         f"  {'metric':<28} {'baseline':>12} {'candidate':>12} {'delta':>14}  status",
     ]
@@ -155,6 +213,23 @@ def render_compare(result: dict) -> str:
             f"  {m['metric']:<28} {fmt(m['baseline']):>12} "
             f"{fmt(m['candidate']):>12} {delta_s:>14}  {m['status']}"
         )
+    per_rank = [
+        r for r in result.get("per_rank_step_time", [])
+        if r["baseline"] is not None or r["candidate"] is not None
+    ]
+    if len(per_rank) > 1:
+        lines.append("")
+        lines.append("  step_time_ms by rank:")
+        for r in per_rank:
+            delta_s = "—"
+            if r["delta"] is not None:
+                sign = "+" if r["delta"] >= 0 else ""
+                delta_s = f"{sign}{r['delta']:.1f}"
+                if r["pct"] is not None:
+                    delta_s += f" ({sign}{r['pct']:.1f}%)"
+            b = "—" if r["baseline"] is None else f"{r['baseline']:.1f}"
+            c = "—" if r["candidate"] is None else f"{r['candidate']:.1f}"
+            lines.append(f"    r{r['rank']:<4} {b:>10} -> {c:>10}  {delta_s}")
     return "\n".join(lines)
 
 
