@@ -113,3 +113,29 @@ def test_crash_stderr_tail_written(tmp_path):
     assert "deliberate crash for tail test" in tail_log.read_text()
     manifest = json.loads((session / "manifest.json").read_text())
     assert manifest["status"] == "failed"
+
+
+@pytest.mark.timeout(240)
+def test_watch_mode_cli_display(tmp_path):
+    """`traceml-amd watch` = run with the live Rich CLI display; must work
+    headless (no tty) and still finalize."""
+    script = tmp_path / "train_tiny.py"
+    script.write_text(SCRIPT)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "30"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "watch",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "watch",
+            "--aggregator-port", "29879",
+            "--master-port", "29603",
+            str(script),
+        ],
+        env=env, capture_output=True, text=True, timeout=220, cwd=REPO_ROOT,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    summary = tmp_path / "logs" / "watch" / "final_summary.json"
+    assert summary.exists()
