@@ -139,3 +139,54 @@ def test_watch_mode_cli_display(tmp_path):
     assert proc.returncode == 0, proc.stderr[-3000:]
     summary = tmp_path / "logs" / "watch" / "final_summary.json"
     assert summary.exists()
+
+
+@pytest.mark.timeout(240)
+def test_dashboard_mode_serves_live_api(tmp_path):
+    """mode=dashboard: the aggregator's FastAPI app serves /api/live while
+    training runs and still finalizes on exit."""
+    import time
+    import urllib.request
+
+    script = tmp_path / "train_tiny.py"
+    script.write_text(SCRIPT.replace("if n >= 60:", "if n >= 200:"))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "30"
+    env["TRACEML_DASHBOARD_PORT"] = "28791"
+    env["TRACEML_INTERVAL"] = "0.5"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--mode", "dashboard",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "dash",
+            "--aggregator-port", "29880",
+            "--master-port", "29604",
+            str(script),
+        ],
+        env=env, cwd=REPO_ROOT,
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+    )
+    got_api = False
+    try:
+        deadline = time.time() + 120
+        while time.time() < deadline and proc.poll() is None:
+            try:
+                with urllib.request.urlopen(
+                    "http://127.0.0.1:28791/api/live", timeout=2
+                ) as r:
+                    payload = json.loads(r.read())
+                if payload.get("step_time", {}).get("steps_analyzed"):
+                    got_api = True
+                    break
+            except Exception:
+                time.sleep(0.5)
+        out, err = proc.communicate(timeout=150)
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+    assert got_api, "dashboard /api/live never served data"
+    assert proc.returncode == 0, err[-2000:]
+    assert (tmp_path / "logs" / "dash" / "final_summary.json").exists()

@@ -70,8 +70,10 @@ __global__ __launch_bounds__(64, 1) void stamp_kernel(Slot* __restrict__ ring,
     staged.seq = seq;
     Slot* slot = ring + (seq & mask);
     slot->ticks = staged.ticks;
-    // Publish: ticks must be host-visible before seq flips to the new value.
-    __threadfence_system();
+    // Publish: the system-scope RELEASE store orders the ticks write before
+    // the seq flip (C++ release semantics cover all prior stores), so no
+    // separate __threadfence_system() is needed — saving ~1 us of fence
+    // drain per stamp on the hot path.
     __hip_atomic_store(&slot->seq, staged.seq, __ATOMIC_RELEASE,
                        __HIP_MEMORY_SCOPE_SYSTEM);
   }
