@@ -158,3 +158,13 @@ def test_aggregate_median_and_worst_across_ranks():
     assert window.median["forward_ms"]["idx"] == 1
     assert window.worst["forward_ms"]["value"] == pytest.approx(90.0)
     assert window.worst["forward_ms"]["idx"] == 2
+
+
+def test_cohorts_partition_ranks():
+    rows = []
+    for rank, fwd in ((0, 30.0), (1, 30.0), (2, 90.0), (3, 10.0)):
+        rows += [_row(rank, s, _events(fwd=fwd)) for s in (1, 2, 3)]
+    window = StepTimeAnalyzer().analyze(rows)
+    assert set(window.cohorts["slow"]) == {2}
+    assert set(window.cohorts["fast"]) == {3}
+    assert set(window.cohorts["typical"]) == {0, 1}

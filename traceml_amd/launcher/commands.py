@@ -79,7 +79,30 @@ def launch_process(
 
     agg_config = AggregatorLaunchConfig(node_rank=node_rank)
     agg_proc = None
+    train_proc = None
     telemetry_status = "ok"
+
+    # Children run in their own sessions (process groups), so a Ctrl-C on
+    # the launcher would otherwise orphan them: forward the signal to both
+    # groups and let the normal monitor/finalize path run
+    # (reference: launcher/process.py:283 signal handlers).
+    import signal as _signal
+
+    def _forward(signum, frame):
+        for child in (train_proc, agg_proc):
+            if child is not None and child.poll() is None:
+                try:
+                    os.killpg(child.pid, _signal.SIGTERM)
+                except (ProcessLookupError, PermissionError):
+                    pass
+
+    previous_handlers = {}
+    for sig in (_signal.SIGINT, _signal.SIGTERM):
+        try:
+            previous_handlers[sig] = _signal.signal(sig, _forward)
+        except (ValueError, OSError):
+            pass  # not the main thread / unsupported
+
     try:
         if agg_config.is_owner:
             agg_proc, _ = spawn_process_group(
@@ -138,6 +161,11 @@ def launch_process(
             terminate_process_group(
                 agg_proc, grace_sec=float(settings.finalize_timeout)
             )
+        for sig, handler in previous_handlers.items():
+            try:
+                _signal.signal(sig, handler)
+            except (ValueError, OSError):
+                pass
 
     summary_path = protocol.summary_json_path(sdir)
     summary_ok = os.path.exists(summary_path)

@@ -62,6 +62,7 @@ def build(db_path: str) -> dict:
         "diagnosis_clock": window.clock,
         "signal_coverage": window.signal_coverage,
         "shares": window.shares,
+        "cohorts": window.cohorts,
         "training_strategy": window.training_strategy,
     }
     payload["card"] = _card(window, payload)

@@ -106,6 +106,7 @@ class StepTimeAnalyzer:
 
         self._aggregate(window)
         self._shares(window)
+        self._cohorts(window)
         return window
 
     # -- stages -------------------------------------------------------------
@@ -288,6 +289,27 @@ class StepTimeAnalyzer:
             worst_rank, worst_value = max(measured, key=lambda p: (p[1], -p[0]))
             window.median[metric] = {"value": med_value, "idx": med_rank}
             window.worst[metric] = {"value": worst_value, "idx": worst_rank}
+
+    @staticmethod
+    def _cohorts(window: StepTimeWindow, tolerance: float = 0.10) -> None:
+        """Group ranks by selected step time relative to the median: within
+        ±tolerance -> typical, above -> slow, below -> fast (reference:
+        analysis.py cohorts, :568-655)."""
+        med = window.median.get("step_time_ms")
+        cohorts = {"typical": [], "slow": [], "fast": []}
+        if med and med.get("value"):
+            med_value = med["value"]
+            for rank in window.ranks_used:
+                value = window.ranks[rank].get("step_time_ms")
+                if value is None:
+                    continue
+                if value > med_value * (1 + tolerance):
+                    cohorts["slow"].append(rank)
+                elif value < med_value * (1 - tolerance):
+                    cohorts["fast"].append(rank)
+                else:
+                    cohorts["typical"].append(rank)
+        window.cohorts = cohorts
 
     @staticmethod
     def _shares(window: StepTimeWindow) -> None:

@@ -138,6 +138,8 @@ class StepTimeWindow:
     shares: Dict[str, Optional[float]] = field(default_factory=dict)
     #: rank -> [(step, step_time_ms), ...] for trend detection
     step_series: Dict[int, List] = field(default_factory=dict)
+    #: behavior cohorts: {"typical": [ranks], "slow": [...], "fast": [...]}
+    cohorts: Dict[str, List[int]] = field(default_factory=dict)
 
     @property
     def ranks_used(self) -> List[int]:
