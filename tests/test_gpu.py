@@ -332,3 +332,21 @@ def test_ring_wrap_small_ring_subprocess():
     )
     assert proc.returncode == 0, proc.stderr[-2000:]
     assert "WRAP_OK" in proc.stdout
+
+
+@requires_gpu
+def test_deep_profile_gpu_clock():
+    from traceml_amd.sdk.deep_profile import deep_profile
+
+    model = torch.nn.Sequential(
+        torch.nn.Linear(1024, 4096), torch.nn.ReLU(), torch.nn.Linear(4096, 64)
+    ).cuda()
+    x = torch.randn(256, 1024, device="cuda")
+    with deep_profile(model) as prof:
+        for _ in range(5):
+            model(x)
+    report = prof.report()
+    assert report["clock"] == "gpu"
+    by_name = {r["module"]: r for r in report["modules"]}
+    assert by_name["0"]["gpu_ms"] is not None and by_name["0"]["gpu_ms"] > 0
+    assert by_name["0"]["gpu_ms"] > by_name["1"]["gpu_ms"]  # big GEMM > relu
