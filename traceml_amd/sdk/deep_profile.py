@@ -94,6 +94,15 @@ class DeepProfile:
                 continue
             self._handles.append(module.register_forward_pre_hook(self._pre(name)))
             self._handles.append(module.register_forward_hook(self._post(name)))
+        if not self._handles:
+            # a bare leaf model (no named submodules): profile the root
+            name = type(self.model).__name__
+            self._handles.append(
+                self.model.register_forward_pre_hook(self._pre(name))
+            )
+            self._handles.append(
+                self.model.register_forward_hook(self._post(name))
+            )
         return self
 
     def __exit__(self, *exc):
