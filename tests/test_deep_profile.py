@@ -35,3 +35,16 @@ def test_deep_profile_hooks_removed_on_exit():
     model(torch.randn(2, 8))  # outside: must not record
     assert prof.report()["modules_profiled"] == before
     assert prof.report()["modules"][0]["calls"] == 1
+
+
+def test_deep_profile_backward():
+    model = nn.Sequential(nn.Linear(128, 512), nn.ReLU(), nn.Linear(512, 8))
+    x = torch.randn(32, 128)
+    with deep_profile(model, backward=True) as prof:
+        for _ in range(3):
+            model(x).sum().backward()
+    report = prof.report(top_k=20)
+    names = {r["module"] for r in report["modules"]}
+    assert "0" in names and "0 [bwd]" in names
+    by_name = {r["module"]: r for r in report["modules"]}
+    assert by_name["0 [bwd]"]["calls"] == 3

@@ -36,9 +36,12 @@ class _ModuleRecord:
 
 
 class DeepProfile:
-    def __init__(self, model, leaf_only: bool = True) -> None:
+    def __init__(
+        self, model, leaf_only: bool = True, backward: bool = False
+    ) -> None:
         self.model = model
         self.leaf_only = leaf_only
+        self.backward = backward
         self._records: Dict[str, _ModuleRecord] = {}
         self._handles: List = []
         self._backend = None
@@ -86,23 +89,50 @@ class DeepProfile:
 
     # -- lifecycle -------------------------------------------------------------
 
+    def _attach(self, name: str, module) -> None:
+        self._handles.append(module.register_forward_pre_hook(self._pre(name)))
+        self._handles.append(module.register_forward_hook(self._post(name)))
+        if self.backward:
+            # backward timing per module: hooks fire in reverse topological
+            # order during autograd; the same pre/post bracket applies
+            bwd_name = name + " [bwd]"
+            self._handles.append(
+                module.register_full_backward_pre_hook(
+                    self._backward_pre(bwd_name)
+                )
+            )
+            self._handles.append(
+                module.register_full_backward_hook(self._backward_post(bwd_name))
+            )
+
+    def _backward_pre(self, name: str):
+        pre = self._pre(name)
+
+        def hook(module, grad_output):
+            pre(module, grad_output)
+            return None
+
+        return hook
+
+    def _backward_post(self, name: str):
+        post = self._post(name)
+
+        def hook(module, grad_input, grad_output):
+            post(module, grad_input, grad_output)
+            return None
+
+        return hook
+
     def __enter__(self) -> "DeepProfile":
         for name, module in self.model.named_modules():
             if not name:
                 continue
             if self.leaf_only and any(module.children()):
                 continue
-            self._handles.append(module.register_forward_pre_hook(self._pre(name)))
-            self._handles.append(module.register_forward_hook(self._post(name)))
+            self._attach(name, module)
         if not self._handles:
             # a bare leaf model (no named submodules): profile the root
-            name = type(self.model).__name__
-            self._handles.append(
-                self.model.register_forward_pre_hook(self._pre(name))
-            )
-            self._handles.append(
-                self.model.register_forward_hook(self._post(name))
-            )
+            self._attach(type(self.model).__name__, self.model)
         return self
 
     def __exit__(self, *exc):
@@ -161,8 +191,10 @@ class DeepProfile:
         }
 
 
-def deep_profile(model, leaf_only: bool = True) -> DeepProfile:
-    return DeepProfile(model, leaf_only=leaf_only)
+def deep_profile(
+    model, leaf_only: bool = True, backward: bool = False
+) -> DeepProfile:
+    return DeepProfile(model, leaf_only=leaf_only, backward=backward)
 
 
 def render_report(report: dict) -> str:
