@@ -124,6 +124,52 @@ class SQLiteStepTimeRepository:
     def load_live(self) -> List[StepTimeSourceRow]:
         return self._load(LIVE_WINDOW_ROWS)
 
+    def load_new_rows(self, since_id: int) -> tuple:
+        """Incremental live read: rows with id > since_id, ordered by id.
+        Returns (rows, max_id_seen). Powers the live session's cursor reuse
+        so a UI tick costs only the new rows, not a full window re-read."""
+        try:
+            conn = _connect(self.db_path)
+        except sqlite3.Error:
+            return ([], since_id)
+        try:
+            try:
+                raw_rows = conn.execute(
+                    """
+                    SELECT id, global_rank, local_rank, node_rank, hostname,
+                           world_size, local_world_size, timestamp, step,
+                           events_json
+                    FROM step_time_samples WHERE id > ? ORDER BY id
+                    """,
+                    (since_id,),
+                ).fetchall()
+            except sqlite3.Error:
+                return ([], since_id)
+        finally:
+            conn.close()
+        rows: List[StepTimeSourceRow] = []
+        max_id = since_id
+        for r in raw_rows:
+            max_id = max(max_id, int(r["id"]))
+            events = normalize_step_time_events(r["events_json"])
+            if events is None or r["step"] is None or r["global_rank"] is None:
+                continue
+            rows.append(
+                StepTimeSourceRow(
+                    row_id=int(r["id"]),
+                    global_rank=int(r["global_rank"]),
+                    step=int(r["step"]),
+                    timestamp=float(r["timestamp"] or 0.0),
+                    events=events,
+                    node_rank=r["node_rank"],
+                    local_rank=r["local_rank"],
+                    hostname=r["hostname"],
+                    world_size=r["world_size"],
+                    local_world_size=r["local_world_size"],
+                )
+            )
+        return (rows, max_id)
+
     def load_summary(self) -> List[StepTimeSourceRow]:
         return self._load(SUMMARY_WINDOW_ROWS)
 
