@@ -46,11 +46,13 @@ def normalize_step_time_events(raw: Optional[str]) -> Optional[dict]:
         duration = cell.get("duration_ms")
         cpu = cell.get("cpu_ms")
         gpu = cell.get("gpu_ms")
-        if isinstance(duration, str):
+        # fast path: our own writer emits numbers/None; anything else
+        # (strings, lists, bools from a hostile peer) goes through _num
+        if duration is not None and type(duration) not in (int, float):
             duration = _num(duration)
-        if isinstance(cpu, str):
+        if cpu is not None and type(cpu) not in (int, float):
             cpu = _num(cpu)
-        if isinstance(gpu, str):
+        if gpu is not None and type(gpu) not in (int, float):
             gpu = _num(gpu)
         events[signal] = {
             "duration_ms": duration,
