@@ -39,7 +39,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--model", default="resnet50", choices=["resnet50", "mlp"])
-    p.add_argument("--batch", type=int, default=64, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=256, help="per-GPU batch size")
     return p.parse_args()
 
 
