@@ -218,13 +218,6 @@ def drain_step_time_queue(max_batches: Optional[int] = None) -> List[StepTimeBat
     return out
 
 
-def requeue_front(batches: List[StepTimeBatch]) -> None:
-    """Sampler-side: put unresolved batches back at the FRONT, oldest first."""
-    with _queue_lock:
-        for batch in reversed(batches):
-            _step_time_queue.appendleft(batch)
-
-
 def queue_depth() -> int:
     with _queue_lock:
         return len(_step_time_queue)
