@@ -39,6 +39,7 @@ _PUBLIC = {
     "wrap_backward": ("traceml_amd.api", "wrap_backward"),
     "wrap_optimizer": ("traceml_amd.api", "wrap_optimizer"),
     "wrap_h2d": ("traceml_amd.api", "wrap_h2d"),
+    "deep_profile": ("traceml_amd.sdk.deep_profile", "deep_profile"),
 }
 
 __all__ = ["__version__", *list(_PUBLIC)]
