@@ -134,12 +134,50 @@ def build_verdict_text(payload: dict) -> str:
                 f"[{diag.get('severity', 'info')}]"
             )
     lines.append("")
+    rank_table = _rank_evidence_table(payload)
+    if rank_table:
+        lines.append(rank_table)
+        lines.append("")
     for name in ("step_time", "step_memory", "system", "process"):
         card = payload.get(name, {}).get("card")
         if card:
             lines.append(card)
             lines.append("")
     return "\n".join(lines).rstrip() + "\n"
+
+
+def _rank_evidence_table(payload: dict) -> str:
+    """Compact per-rank step-time evidence (multi-rank runs only)."""
+    rows = payload.get("step_time", {}).get("groups", {}).get("rows", {})
+    if len(rows) < 2:
+        return ""
+    metrics = [
+        ("step_time_ms", "step"),
+        ("input_wait_ms", "input"),
+        ("forward_ms", "fwd"),
+        ("backward_ms", "bwd"),
+        ("optimizer_ms", "opt"),
+        ("ddp_comm_ms", "comm"),
+        ("residual_ms", "resid"),
+    ]
+    keys = sorted(rows, key=lambda k: (len(k), k))
+    used = [
+        (m, label)
+        for m, label in metrics
+        if any(rows[k]["metrics"].get(m) is not None for k in keys)
+    ]
+    if not used:
+        return ""
+    header = "  rank   " + "".join(f"{label:>9}" for _, label in used) + "   (ms)"
+    lines = ["Per-rank step time:", header]
+    for key in keys:
+        cells = rows[key]["metrics"]
+        line = f"  r{key:<6}"
+        for m, _label in used:
+            value = cells.get(m)
+            line += f"{value:>9.1f}" if value is not None else f"{'—':>9}"
+        lines.append(line)
+    return "\n".join(lines)
 
 
 def write_summary_artifacts(
