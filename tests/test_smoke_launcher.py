@@ -190,3 +190,38 @@ def test_dashboard_mode_serves_live_api(tmp_path):
     assert got_api, "dashboard /api/live never served data"
     assert proc.returncode == 0, err[-2000:]
     assert (tmp_path / "logs" / "dash" / "final_summary.json").exists()
+
+
+@pytest.mark.timeout(400)
+def test_four_rank_input_straggler_e2e(tmp_path):
+    """The flagship diagnosis end-to-end: 4 DDP ranks (gloo), rank 2 slow
+    dataloader -> INPUT STRAGGLER [crit] with culprit r2 and measured
+    ddp_comm evidence (BASELINE config 3 on CPU)."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "40"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--nproc-per-node", "4",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "strag",
+            "--aggregator-port", "29882",
+            "--master-port", "29605",
+            os.path.join(REPO_ROOT, "examples", "demo",
+                         "mlp_ddp_input_straggler.py"),
+        ],
+        env=env, capture_output=True, text=True, timeout=380, cwd=REPO_ROOT,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    payload = json.loads(
+        (tmp_path / "logs" / "strag" / "final_summary.json").read_text()
+    )
+    primary = payload["primary_diagnosis"]
+    assert primary["kind"] == "INPUT_STRAGGLER"
+    assert primary["severity"] == "crit"
+    st_diag = payload["step_time"]["diagnosis"]
+    assert st_diag["ranks"] == [2]
+    assert "ddp_comm_ms_per_rank" in st_diag["evidence"]
+    assert payload["step_time"]["metadata"]["global_ranks_used"] == [0, 1, 2, 3]
