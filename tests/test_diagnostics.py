@@ -2,7 +2,6 @@
 the cross-surface contract suite (mirrors reference
 tests/step_time/test_contract_baseline.py + tests/diagnostics/*)."""
 
-import os
 
 import pytest
 

@@ -227,3 +227,42 @@ def test_export_chrome_trace(tmp_path, capsys):
     r0_input = [e["dur"] for e in phase_events
                 if e["pid"] == 0 and e["name"] == "input_wait"]
     assert min(r2_input) > 10 * max(r0_input)
+
+
+def test_trend_never_primary_over_other_findings(tmp_path):
+    """A degrading-but-compute-bound window must diagnose COMPUTE_BOUND
+    with STEP_TIME_DEGRADING as a supporting issue (never issues[0])."""
+    import json as _json
+    import sqlite3 as _sqlite3
+    import time as _time
+
+    from tests import scenarios
+    from traceml_amd.aggregator.writers import build_all_writers
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    db = str(tmp_path / "t.sqlite")
+    conn = _sqlite3.connect(db)
+    for w in build_all_writers():
+        w.init_schema(conn)
+    with conn:
+        for step in range(1, 61):
+            scale = 1.0 + step / 60.0  # 2x degradation across the window
+            profile = scenarios.RankProfile(
+                input_ms=0.2,
+                h2d_ms=0.1,
+                forward_ms=30.0 * scale,
+                backward_ms=55.0 * scale,
+                optimizer_ms=5.0 * scale,
+            )
+            conn.execute(
+                "INSERT INTO step_time_samples (global_rank, world_size, "
+                "timestamp, step, events_json) VALUES (0, 1, ?, ?, ?)",
+                (_time.time() + step * 0.1, step,
+                 _json.dumps(profile.events())),
+            )
+    conn.close()
+    result = StepTimePipeline(db, profile="summary").run()
+    kinds = [i.kind for i in result.diagnosis.issues]
+    assert "STEP_TIME_DEGRADING" in kinds
+    assert result.diagnosis.primary.kind != "STEP_TIME_DEGRADING"
+    assert result.diagnosis.primary.kind == "COMPUTE_BOUND"

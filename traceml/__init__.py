@@ -22,6 +22,3 @@ def __getattr__(name):
     return getattr(_impl, name)
 
 
-# Let `import traceml.api`-style submodule imports resolve to traceml_amd's.
-for _sub in list(sys.modules):
-    pass

@@ -228,7 +228,8 @@ def evaluate(window: StepTimeWindow) -> List[DiagnosticIssue]:
     if trend_issue is not None:
         issues.append(trend_issue)
 
-    if not [i for i in issues if i.kind != "STEP_TIME_DEGRADING"]:
+    non_trend = [i for i in issues if i.kind != "STEP_TIME_DEGRADING"]
+    if not non_trend:
         issues.insert(
             0,
             DiagnosticIssue(
@@ -243,7 +244,13 @@ def evaluate(window: StepTimeWindow) -> List[DiagnosticIssue]:
                 },
             )
         )
-    return sort_issues(issues)
+    ordered = sort_issues(issues)
+    # The degradation trend is supporting evidence, never THE diagnosis
+    # when any other finding (including BALANCED/COMPUTE_BOUND) exists.
+    if len(ordered) > 1 and ordered[0].kind == "STEP_TIME_DEGRADING":
+        trend = ordered.pop(0)
+        ordered.insert(1, trend)
+    return ordered
 
 
 def _trend_issue(window: StepTimeWindow):
