@@ -81,6 +81,15 @@ class StepTimeSampler(BaseSampler):
                 if now - batch.flushed_at < STALE_BATCH_SEC:
                     break  # ordered: retry this and everything behind next tick
                 # Stale: abandon the GPU side for the stuck events.
+                import logging
+
+                logging.getLogger(__name__).warning(
+                    "traceml_amd: step %d GPU stamps unresolved after %.0fs "
+                    "(%s) — shipping CPU-only",
+                    batch.step,
+                    now - batch.flushed_at,
+                    ",".join(e.name.split(":")[-1] for e in unresolved),
+                )
                 for event in unresolved:
                     event.gpu_start = event.gpu_end = None
                     event._gpu_done = True
