@@ -225,3 +225,31 @@ def test_four_rank_input_straggler_e2e(tmp_path):
     assert st_diag["ranks"] == [2]
     assert "ddp_comm_ms_per_rank" in st_diag["evidence"]
     assert payload["step_time"]["metadata"]["global_ranks_used"] == [0, 1, 2, 3]
+
+
+@pytest.mark.timeout(240)
+def test_trace_max_steps_budget_e2e(tmp_path):
+    """--trace-max-steps caps the recorded window through the full stack."""
+    script = tmp_path / "train_tiny.py"
+    script.write_text(SCRIPT)  # runs 60 steps
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "30"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "budget",
+            "--aggregator-port", "29885",
+            "--master-port", "29606",
+            "--trace-max-steps", "25",
+            str(script),
+        ],
+        env=env, capture_output=True, text=True, timeout=220, cwd=REPO_ROOT,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    payload = json.loads(
+        (tmp_path / "logs" / "budget" / "final_summary.json").read_text()
+    )
+    assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 25

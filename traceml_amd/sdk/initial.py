@@ -229,6 +229,10 @@ def init(
             return config
 
         _apply_requested_patches(config)
+        if settings.trace_max_steps:
+            from traceml_amd.runtime import state
+
+            state.recording_state().set_max_steps(settings.trace_max_steps)
         _active_config = config
         return config
 
