@@ -22,7 +22,6 @@ WORKER = textwrap.dedent(
 
     sys.path.insert(0, os.environ["TRACEML_AMD_REPO"])
     from traceml_amd.core import timing, event_names
-    from traceml_amd.parallel.ddp_hook import attach_ddp_comm_timing
     from traceml_amd.parallel.rank_stats import (
         enable_rank_stats_exchange, FIELDS,
     )
@@ -37,8 +36,9 @@ WORKER = textwrap.dedent(
     initial._apply_requested_patches(config)
     initial._active_config = config
 
+    # NOTE: no manual attach — trace_step auto-attaches the ddp_comm hook
+    # on the first step of a DDP model (settings.ddp_comm_timing default)
     model = DDP(nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4)))
-    attach_ddp_comm_timing(model)
     exchange = enable_rank_stats_exchange(min_interval_sec=0.0)
     assert exchange is not None
     opt = torch.optim.SGD(model.parameters(), lr=0.01)

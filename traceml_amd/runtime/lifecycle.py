@@ -116,6 +116,16 @@ def start_runtime(
     register_atexit: bool = True,
 ) -> RuntimeHandle:
     settings = settings or TraceMLSettings.from_env()
+    if settings.enable_logging and settings.logs_dir:
+        try:
+            from traceml_amd.runtime.session import get_session_id, session_dir
+            from traceml_amd.utils.loggers import setup_error_logger
+
+            setup_error_logger(
+                session_dir(settings.logs_dir, get_session_id(settings.session_id))
+            )
+        except Exception:
+            pass
     try:
         runtime = TraceMLRuntime(settings)
         runtime.start()

@@ -83,6 +83,7 @@ class _TraceStep:
                 if len(_cached_targets) > 64:
                     _cached_targets.clear()
                 _cached_targets[key] = targets
+                _maybe_attach_ddp_timing(model, config)
             flags.forward_targets = targets
             flags.forward_enabled = config.patch_forward
         else:
@@ -124,6 +125,24 @@ def trace_step(model=None):
     ):
         return _NoopStep(advance=config is not None and not config.noop)
     return _TraceStep(model, config, recording)
+
+
+def _maybe_attach_ddp_timing(model, config) -> None:
+    """First trace_step of a DDP-wrapped model: auto-attach the ddp_comm
+    timing hook (settings.ddp_comm_timing, on by default). Runs before the
+    first backward, which is when DDP still accepts a comm hook."""
+    if not getattr(config.settings, "ddp_comm_timing", True):
+        return
+    try:
+        from torch.nn.parallel import DistributedDataParallel
+
+        if not isinstance(model, DistributedDataParallel):
+            return
+        from traceml_amd.parallel.ddp_hook import attach_ddp_comm_timing
+
+        attach_ddp_comm_timing(model)
+    except Exception:
+        pass  # user may have registered their own comm hook — keep theirs
 
 
 def _kick_rank_stats(step: int) -> None:
