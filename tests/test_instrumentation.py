@@ -129,3 +129,32 @@ def test_optimizer_hook_auto_mode_only(armed_auto_config, tiny_model):
 
     with pytest.raises(RuntimeError):
         wrap_forward(tiny_model)
+
+
+def test_trace_time_user_region(armed_auto_config, tiny_model):
+    """Custom regions: recorded inside the step, survive normalization as
+    user:<name> signals, ignored by the analyzer's derived metrics."""
+    import json
+
+    from traceml_amd.sdk.instrumentation import trace_step, trace_time
+    from traceml_amd.steptime.analyzer import StepTimeAnalyzer
+    from traceml_amd.steptime.model import StepTimeSourceRow
+    from traceml_amd.steptime.repository import normalize_step_time_events
+
+    @trace_time("augmentation")
+    def augment(x):
+        return x * 2
+
+    with trace_step(tiny_model):
+        x = augment(torch.randn(2, 8))
+        tiny_model(x).sum().backward()
+    rows = drain_step_time_rows()
+    assert "_traceml_user:augmentation" in rows[0]["events"]
+
+    events = normalize_step_time_events(json.dumps(rows[0]["events"]))
+    assert "user:augmentation" in events
+    window = StepTimeAnalyzer().analyze(
+        [StepTimeSourceRow(row_id=1, global_rank=0, step=1, timestamp=0.0,
+                           events=events)]
+    )
+    assert window.ranks[0].backward_ms is not None  # derivations unaffected

@@ -41,6 +41,11 @@ def normalize_step_time_events(raw: Optional[str]) -> Optional[dict]:
     names = STEP_TIME_EVENT_NAMES
     for wire_name, cell in decoded.items():
         signal = names.get(wire_name)
+        if signal is None and wire_name.startswith("_traceml_user:"):
+            # custom trace_time regions survive as "user:<name>" signals —
+            # ignored by the analyzer's derivations, visible to inspect /
+            # export-trace / custom tooling
+            signal = "user:" + wire_name.split(":", 1)[1]
         if signal is None or type(cell) is not dict:
             continue
         duration = cell.get("duration_ms")
