@@ -102,3 +102,85 @@ def test_serve_and_direct_init(tmp_path):
     assert summary_path.exists()
     payload = json.loads(summary_path.read_text())
     assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 40
+
+
+MIDRUN_SCRIPT = textwrap.dedent(
+    """
+    import sys, os
+    sys.path.insert(0, os.environ["REPO"])
+    import torch, torch.nn as nn
+    import traceml_amd
+
+    traceml_amd.init(
+        aggregator_port=int(os.environ["PORT"]),
+        logs_dir=os.environ["LOGS"],
+        session_id="midrun",
+        connect_timeout_sec=20.0,
+    )
+    model = nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    for _ in range(30):
+        with traceml_amd.trace_step(model):
+            opt.zero_grad()
+            model(torch.randn(8, 16)).sum().backward()
+            opt.step()
+    import time
+    time.sleep(2.5)  # let the tail ship
+    flat = traceml_amd.summary(timeout_sec=45.0)   # in-run file-RPC
+    assert flat.get("traceml/verdict_kind"), flat
+    assert isinstance(flat.get("traceml/step_time/step_time_ms"), float), flat
+    full = traceml_amd.final_summary(wait=True, timeout_sec=30.0)
+    assert full["schema_version"] == 1.7
+    print("MIDRUN_OK", flat["traceml/verdict_kind"])
+    """
+)
+
+
+@pytest.mark.timeout(300)
+def test_midrun_summary_rpc(tmp_path):
+    """traceml_amd.summary() during the run: file-RPC to the live
+    aggregator produces a flat tracker dict before the run ends."""
+    port = "29892"
+    env = dict(os.environ)
+    env.update(
+        {
+            "REPO": REPO_ROOT,
+            "PORT": port,
+            "LOGS": str(tmp_path / "logs"),
+            "PYTHONPATH": REPO_ROOT + os.pathsep + env.get("PYTHONPATH", ""),
+            "TRACEML_FINALIZE_TIMEOUT": "25",
+            "TRACEML_INTERVAL": "0.5",
+        }
+    )
+    serve = subprocess.Popen(
+        [
+            sys.executable, "-m", "traceml_amd", "serve",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "midrun",
+            "--aggregator-port", port,
+            "--interval", "0.5",
+        ],
+        env=env,
+    )
+    try:
+        from traceml_amd.transport.tcp import probe_tcp
+
+        deadline = time.time() + 60
+        while time.time() < deadline and not probe_tcp("127.0.0.1", int(port)):
+            time.sleep(0.25)
+            assert serve.poll() is None
+        script = tmp_path / "user.py"
+        script.write_text(MIDRUN_SCRIPT)
+        user = subprocess.run(
+            [sys.executable, str(script)], env=env, capture_output=True,
+            text=True, timeout=200,
+        )
+        assert user.returncode == 0, (user.stdout[-1500:], user.stderr[-1500:])
+        assert "MIDRUN_OK" in user.stdout
+    finally:
+        serve.send_signal(signal.SIGTERM)
+        try:
+            serve.wait(timeout=60)
+        except subprocess.TimeoutExpired:
+            serve.kill()
+            raise

@@ -23,7 +23,14 @@ from traceml_amd.utils.atomic_io import atomic_write_json
 
 
 def _session_dir() -> str:
-    settings = TraceMLSettings.from_env()
+    # the settings the process was init()ed with win over raw env (a direct
+    # init(logs_dir=..., session_id=...) launch has no TRACEML_* env set)
+    from traceml_amd.sdk import initial
+
+    config = initial.get_active_config()
+    settings = (
+        config.settings if config is not None else TraceMLSettings.from_env()
+    )
     return session_dir(settings.logs_dir, get_session_id(settings.session_id))
 
 
