@@ -73,3 +73,38 @@ def test_cli_driver_renders_without_terminal(db_path):
     text = console.file.getvalue()
     assert "traceml-amd live" in text
     assert "backward" in text
+
+
+def test_summary_pipeline_query_topology(db_path, monkeypatch):
+    """Pin the SQL shape of one summary pipeline run (reference:
+    SQLiteSelectRecorder / test_query_topology): exactly one step_time
+    SELECT + one strategy SELECT, no N+1 patterns."""
+    import sqlite3 as sqlite3_mod
+
+    scenarios.healthy_ddp(ranks=4, steps=20).write(db_path)
+    selects = []
+    original_connect = sqlite3_mod.connect
+
+    class _RecordingConnection:
+        def __init__(self, conn):
+            self._conn = conn
+
+        def execute(self, sql, *args):
+            if sql.strip().upper().startswith("SELECT"):
+                selects.append(" ".join(sql.split()))
+            return self._conn.execute(sql, *args)
+
+        def __getattr__(self, name):
+            return getattr(self._conn, name)
+
+    def connect(*args, **kwargs):
+        return _RecordingConnection(original_connect(*args, **kwargs))
+
+    monkeypatch.setattr(sqlite3_mod, "connect", connect)
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    result = StepTimePipeline(db_path, profile="summary").run()
+    assert result.window.steps_analyzed == 20
+    assert len(selects) == 2, selects
+    assert "step_time_samples" in selects[0]
+    assert "runtime_environment" in selects[1]
