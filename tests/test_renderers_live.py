@@ -87,7 +87,7 @@ def test_summary_pipeline_query_topology(db_path, monkeypatch):
 
     class _RecordingConnection:
         def __init__(self, conn):
-            self._conn = conn
+            object.__setattr__(self, "_conn", conn)
 
         def execute(self, sql, *args):
             if sql.strip().upper().startswith("SELECT"):
@@ -95,7 +95,10 @@ def test_summary_pipeline_query_topology(db_path, monkeypatch):
             return self._conn.execute(sql, *args)
 
         def __getattr__(self, name):
-            return getattr(self._conn, name)
+            return getattr(object.__getattribute__(self, "_conn"), name)
+
+        def __setattr__(self, name, value):  # e.g. row_factory
+            setattr(object.__getattribute__(self, "_conn"), name, value)
 
     def connect(*args, **kwargs):
         return _RecordingConnection(original_connect(*args, **kwargs))
