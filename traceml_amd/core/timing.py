@@ -100,6 +100,9 @@ _dropped_batches = 0
 def _gpu_mark_if_enabled(name: str) -> Optional[int]:
     if name in _CPU_ONLY_EVENTS:
         return None
+    # A GPU machine without the native extension must fail LOUDLY (policy:
+    # a silent CPU-clock fallback could masquerade as the native path), so
+    # GpuTimerUnavailable propagates; only transient mark errors degrade.
     backend = gpu_timer.get_backend()
     if backend is None:
         return None
