@@ -112,3 +112,30 @@ def test_summary_projection_flat_dict(db_path):
     assert all(
         isinstance(v, (int, float, str, type(None))) for v in flat.values()
     )
+
+
+def test_input_bound_action_names_dataloader_workers(db_path, tmp_path):
+    """When the code manifest shows num_workers=0 and the verdict is
+    INPUT_BOUND, the action cites the exact construction site."""
+    from traceml_amd.utils.atomic_io import atomic_write_json
+
+    scenarios.input_bound(steps=30).write(db_path)
+    atomic_write_json(
+        str(tmp_path / "code_manifest.json"),
+        {
+            "calls": [
+                {"call": "DataLoader", "line": 42,
+                 "kwargs": {"num_workers": 0, "batch_size": 32}}
+            ]
+        },
+    )
+    payload = generate_summary(db_path, str(tmp_path))
+    action = payload["primary_diagnosis"]["action"]
+    assert "num_workers=0" in action and "line 42" in action
+    assert "num_workers=0" in payload["text"]
+
+
+def test_no_manifest_no_hint(db_path, tmp_path):
+    scenarios.input_bound(steps=30).write(db_path)
+    payload = generate_summary(db_path, str(tmp_path))
+    assert "num_workers=" not in payload["primary_diagnosis"]["action"]

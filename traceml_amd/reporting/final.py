@@ -199,6 +199,41 @@ def write_summary_artifacts(
     return paths
 
 
+def _enrich_actions_from_code_manifest(payload: dict, session_dir: str) -> None:
+    """Cross-reference the AST code manifest: when the verdict is
+    input-related and the user's DataLoader was constructed with
+    num_workers=0 (or unset), say so in the action — the most common cause,
+    named concretely."""
+    primary = payload.get("primary_diagnosis", {})
+    if primary.get("kind") not in ("INPUT_BOUND", "INPUT_STRAGGLER"):
+        return
+    import json as _json
+
+    try:
+        with open(
+            os.path.join(session_dir, "code_manifest.json"), encoding="utf-8"
+        ) as f:
+            manifest = _json.load(f)
+    except (OSError, ValueError):
+        return
+    for call in manifest.get("calls", []):
+        if call.get("call") != "DataLoader":
+            continue
+        workers = call.get("kwargs", {}).get("num_workers", 0)
+        if workers in (0, None, "<dynamic>") and workers != "<dynamic>":
+            hint = (
+                f" Your script constructs DataLoader with num_workers="
+                f"{workers} (line {call.get('line')}): loading happens on "
+                "the training thread — start with num_workers=8 and "
+                "pin_memory=True."
+            )
+            primary["action"] = primary.get("action", "") + hint
+            st_diag = payload.get("step_time", {}).get("diagnosis")
+            if st_diag:
+                st_diag["action"] = st_diag.get("action", "") + hint
+            break
+
+
 def generate_summary(
     db_path: str,
     session_dir: str,
@@ -206,5 +241,7 @@ def generate_summary(
     html: bool = False,
 ) -> dict:
     payload = FinalReportGenerator(db_path, run_name=run_name).generate()
+    _enrich_actions_from_code_manifest(payload, session_dir)
+    payload["text"] = build_verdict_text(payload)  # re-render with the hint
     write_summary_artifacts(payload, session_dir, html=html)
     return payload
