@@ -72,6 +72,7 @@ WORKER = textwrap.dedent(
         gathered = rows[-1]["ranks"]
         assert len(gathered) == dist.get_world_size()
         assert set(FIELDS).issubset(set(gathered[0]) - {"rank"})
+        assert rows[-1]["gather_latency_ms"] >= 0.0
         print("GATHERED_OK", json.dumps(gathered[0]["step_ms"]))
     dist.destroy_process_group()
     print("WORKER_OK", rank)

@@ -91,8 +91,9 @@ class RankStatsExchange:
         self._min_interval = min_interval_sec
         self._last_launch = 0.0
         self._lock = threading.Lock()
-        self._inflight = None  # (work, out_tensor, launched_at)
+        self._inflight = None  # (work, launched_at)
         self._gathered: List[dict] = []
+        self._latency_ms: List[float] = []
         self._use_gpu = torch.cuda.is_available() and dist.get_backend() in (
             "nccl",
         )
@@ -149,6 +150,10 @@ class RankStatsExchange:
                 self._inflight = None  # abandon a wedged gather
             return
         self._inflight = None
+        gather_ms = (time.time() - launched_at) * 1000.0
+        self._latency_ms.append(gather_ms)
+        if len(self._latency_ms) > 64:
+            del self._latency_ms[:-64]
         try:
             matrix = self._recv.reshape(self._world, VEC_LEN).cpu().tolist()
         except Exception:
@@ -157,6 +162,11 @@ class RankStatsExchange:
             {
                 "timestamp": time.time(),
                 "world_size": self._world,
+                # wall time from launch to observed completion — upper bound
+                # on the xGMI all-gather latency (poll cadence adds slack)
+                "gather_latency_ms": gather_ms,
+                "gather_latency_ms_mean": sum(self._latency_ms)
+                / len(self._latency_ms),
                 "ranks": [
                     {FIELDS[j]: row[j] for j in range(VEC_LEN)} | {"rank": i}
                     for i, row in enumerate(matrix)
