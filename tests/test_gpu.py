@@ -350,3 +350,55 @@ def test_deep_profile_gpu_clock():
     by_name = {r["module"]: r for r in report["modules"]}
     assert by_name["0"]["gpu_ms"] is not None and by_name["0"]["gpu_ms"] > 0
     assert by_name["0"]["gpu_ms"] > by_name["1"]["gpu_ms"]  # big GEMM > relu
+
+
+@requires_gpu
+def test_ddp_comm_hook_on_rccl_ws1(armed_auto_config):
+    """The full RCCL path of the ddp_comm hook on one GPU (world_size=1):
+    auto-attach, side-stream end stamp, future semantics, gradient
+    correctness — the exact code the 8-GPU scale bench runs per rank."""
+    import os
+
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29631")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from tests.conftest import drain_step_time_rows
+    from traceml_amd.core import event_names
+    from traceml_amd.sdk.instrumentation import trace_step
+
+    torch.manual_seed(7)
+    inner = torch.nn.Linear(32, 16).cuda()
+    reference = torch.nn.Linear(32, 16).cuda()
+    reference.load_state_dict(inner.state_dict())
+    model = DDP(inner, device_ids=[torch.cuda.current_device()])
+
+    x = torch.randn(8, 32, device="cuda")
+    for _ in range(3):
+        with trace_step(model):  # first step auto-attaches the comm hook
+            model(x).sum().backward()
+            model.zero_grad(set_to_none=False)
+    # gradient correctness vs a plain module (ws=1: allreduce is identity)
+    with trace_step(model):
+        model(x).sum().backward()
+    reference(x).sum().backward()
+    torch.cuda.synchronize()
+    assert torch.allclose(
+        inner.weight.grad, reference.weight.grad, rtol=1e-5, atol=1e-6
+    )
+
+    rows = drain_step_time_rows()
+    ddp_cells = [
+        r["events"][event_names.DDP_COMM]
+        for r in rows
+        if event_names.DDP_COMM in r["events"]
+    ]
+    assert ddp_cells, "ddp_comm never recorded on RCCL"
+    assert any(c["gpu_ms"] is not None for c in ddp_cells), (
+        "ddp_comm lost its device clock on RCCL"
+    )
+    dist.destroy_process_group()
