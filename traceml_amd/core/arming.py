@@ -44,3 +44,21 @@ _flags = _PhaseFlags()
 
 def phase_flags() -> _PhaseFlags:
     return _flags
+
+
+# Process-wide step-open indicator (NOT thread-local): consumers that run on
+# other threads — the DDP comm hook fires on the autograd worker thread on
+# GPU — need to know a step bracket is open. Single int mutated under the
+# GIL by the (single) training thread's trace_step enter/exit.
+_steps_open = 0
+
+
+def mark_step_open(opened: bool) -> None:
+    global _steps_open
+    _steps_open += 1 if opened else -1
+    if _steps_open < 0:
+        _steps_open = 0
+
+
+def any_step_open() -> bool:
+    return _steps_open > 0

@@ -24,7 +24,7 @@ import time
 from typing import Optional
 
 from traceml_amd.core import event_names
-from traceml_amd.core.arming import is_tracing_armed, phase_flags
+from traceml_amd.core.arming import any_step_open, is_tracing_armed
 from traceml_amd.core.timing import TimeEvent, record_event
 
 logger = logging.getLogger(__name__)
@@ -42,7 +42,9 @@ class _CommTimerState:
 
 
 def _should_time() -> bool:
-    return is_tracing_armed() and phase_flags().in_step
+    # NOT the thread-local in_step: on GPU this hook runs on the autograd
+    # worker thread, which never sees the training thread's TLS flags.
+    return is_tracing_armed() and any_step_open()
 
 
 def timed_allreduce_hook(state: _CommTimerState, bucket):

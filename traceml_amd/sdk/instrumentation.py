@@ -75,6 +75,9 @@ class _TraceStep:
 
         flags = phase_flags()
         flags.in_step = True
+        from traceml_amd.core import arming as _arming
+
+        _arming.mark_step_open(True)
         if model is not None:
             key = id(model)
             targets = _cached_targets.get(key)
@@ -99,6 +102,9 @@ class _TraceStep:
 
     def __exit__(self, *exc):
         close_event(self._event)
+        from traceml_amd.core import arming as _arming
+
+        _arming.mark_step_open(False)
         flags = phase_flags()
         flags.in_step = False
         flags.forward_enabled = False
