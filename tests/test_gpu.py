@@ -402,3 +402,41 @@ def test_ddp_comm_hook_on_rccl_ws1(armed_auto_config):
         "ddp_comm lost its device clock on RCCL"
     )
     dist.destroy_process_group()
+
+
+@requires_gpu
+def test_rank_stats_gather_on_rccl_ws1():
+    """RCCL mechanics of the rank-stats exchange on hardware (ws=1 group):
+    async all_gather_into_tensor launch, non-blocking is_completed poll,
+    D2H readback — the path every rank runs in the scale bench."""
+    import os
+    import time as _time
+
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29633")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from traceml_amd.core import timing
+    from traceml_amd.parallel.rank_stats import FIELDS, RankStatsExchange
+
+    timing._last_cpu_summary.update(
+        {"step_ms": 13.5, "forward_ms": 4.3, "backward_ms": 8.1}
+    )
+    exchange = RankStatsExchange(min_interval_sec=0.0)
+    assert exchange._use_gpu  # nccl backend + cuda -> GPU tensors
+    exchange.on_step_flushed(7)
+    rows = []
+    deadline = _time.time() + 15
+    while _time.time() < deadline and not rows:
+        _time.sleep(0.05)
+        exchange.on_step_flushed(8)
+        rows = exchange.drain_gathered()
+    assert rows, "RCCL gather never completed"
+    gathered = rows[-1]["ranks"][0]
+    assert gathered["step_ms"] == pytest.approx(13.5)
+    assert set(FIELDS).issubset(set(gathered) - {"rank"})
+    assert rows[-1]["gather_latency_ms"] < 1000.0
+    timing._last_cpu_summary.clear()
+    dist.destroy_process_group()

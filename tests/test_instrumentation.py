@@ -158,3 +158,18 @@ def test_trace_time_user_region(armed_auto_config, tiny_model):
                            events=events)]
     )
     assert window.ranks[0].backward_ms is not None  # derivations unaffected
+
+
+def test_dataloader_timing_with_worker_processes(armed_auto_config, tiny_model):
+    """num_workers>0: the fetch is timed as the parent-side wait on the
+    worker queue (the realistic production configuration)."""
+    ds = TensorDataset(torch.randn(32, 8), torch.randn(32, 4))
+    dl = DataLoader(ds, batch_size=8, num_workers=2)
+    opt = torch.optim.SGD(tiny_model.parameters(), lr=0.1)
+    for x, y in dl:
+        _step_once(tiny_model, x, y, opt)
+    rows = drain_step_time_rows()
+    assert len(rows) == 4
+    for row in rows:
+        assert event_names.DATALOADER in row["events"]
+        assert row["events"][event_names.DATALOADER]["cpu_ms"] >= 0.0
