@@ -91,6 +91,7 @@ class ProcessWriter(ProjectionWriter):
                 ("gpu_mem_reserved_bytes", "INTEGER"),
                 ("gpu_capacity_bytes", "INTEGER"),
                 ("device", "TEXT"),
+                ("traceml_self_overhead_us", "REAL"),
             ],
         )
     }

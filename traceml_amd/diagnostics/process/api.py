@@ -42,7 +42,8 @@ def load_process_context(db_path: str) -> ProcessContext:
                 "MAX(gpu_mem_used_bytes) AS gpu_alloc_max, "
                 "AVG(gpu_mem_reserved_bytes) AS gpu_reserved, "
                 "MAX(gpu_mem_reserved_bytes) AS gpu_reserved_max, "
-                "MAX(gpu_capacity_bytes) AS gpu_capacity "
+                "MAX(gpu_capacity_bytes) AS gpu_capacity, "
+                "AVG(traceml_self_overhead_us) AS self_overhead_us "
                 "FROM process_samples GROUP BY global_rank"
             ).fetchall()
             for r in rows:

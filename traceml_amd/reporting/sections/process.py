@@ -44,6 +44,13 @@ def build(db_path: str) -> dict:
         }
 
     md = payload["metadata"]
+    overheads = [
+        r.get("self_overhead_us")
+        for r in ctx.ranks.values()
+        if r.get("self_overhead_us") is not None
+    ]
+    if overheads:
+        md["traceml_self_overhead_us_per_step"] = sum(overheads) / len(overheads)
     md["mode"] = "single_node" if per_rank else "no_data"
     md["samples"] = sum(int(r.get("n") or 0) for r in ctx.ranks.values()) or None
     md["global_ranks_seen"] = sorted(ctx.ranks)
@@ -68,6 +75,11 @@ def build(db_path: str) -> dict:
                 if reserved is not None
                 else ""
             )
+        )
+    overhead = md.get("traceml_self_overhead_us_per_step")
+    if overhead is not None:
+        lines.append(
+            f"  traceml self-overhead: {overhead:.0f} µs/step (bracket bookkeeping)"
         )
     if diag:
         lines.append(f"  Verdict: {diag.get('status')} — {diag.get('summary')}")

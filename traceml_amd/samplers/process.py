@@ -28,6 +28,15 @@ def _cuda_safe_to_touch() -> bool:
         return False
 
 
+def _self_overhead_us():
+    try:
+        from traceml_amd.sdk.instrumentation import self_overhead_us_per_step
+
+        return self_overhead_us_per_step()
+    except Exception:
+        return None
+
+
 class ProcessSampler(BaseSampler):
     name = "process"
 
@@ -87,5 +96,6 @@ class ProcessSampler(BaseSampler):
                 "gpu_mem_reserved_bytes": gpu_mem_reserved,
                 "gpu_capacity_bytes": gpu_capacity,
                 "device": device,
+                "traceml_self_overhead_us": _self_overhead_us(),
             },
         )

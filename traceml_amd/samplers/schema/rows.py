@@ -37,6 +37,7 @@ class GPUMetricsRow:
 @dataclass
 class ProcessSampleRow:
     timestamp: float
+    traceml_self_overhead_us: Optional[float] = None
     cpu_percent: Optional[float] = None
     cpu_capacity_percent: Optional[float] = None
     ram_bytes: Optional[int] = None

@@ -258,6 +258,9 @@ def reset_for_tests() -> None:
 
         _sdk_instr._cached_identity = None
         _sdk_instr._cached_mem_tracker = None
+        _sdk_instr._cached_targets.clear()
+        _sdk_instr._self_cost_sec = 0.0
+        _sdk_instr._self_steps = 0
         state.session_state().reset_for_tests()
         state.recording_state().reset_for_tests()
         _active_config = None

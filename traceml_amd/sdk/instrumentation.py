@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import functools
 import time
+import time as _time
 
 
 from traceml_amd.core import event_names
@@ -37,6 +38,11 @@ _cached_identity = None
 _cached_mem_tracker = None
 _cached_targets: dict = {}
 
+# self-overhead accounting: time spent inside the bracket's own enter/exit
+# (not user code), published via the process sampler
+_self_cost_sec = 0.0
+_self_steps = 0
+
 
 class _NoopStep:
     __slots__ = ("_advance",)
@@ -54,7 +60,7 @@ class _NoopStep:
 
 
 class _TraceStep:
-    __slots__ = ("_model", "_config", "_recording", "_mem", "_event")
+    __slots__ = ("_model", "_config", "_recording", "_mem", "_event", "_t_enter")
 
     def __init__(self, model, config, recording) -> None:
         self._model = model
@@ -63,6 +69,7 @@ class _TraceStep:
 
     def __enter__(self):
         global _cached_identity, _cached_mem_tracker
+        self._t_enter = _time.perf_counter()
         model = self._model
         config = self._config
         if _cached_identity is None:
@@ -98,9 +105,12 @@ class _TraceStep:
         if config.auto_optimizer_hooks:
             ensure_optimizer_timing_installed()
         self._event = open_event(event_names.STEP_TIME)
+        global _self_cost_sec
+        _self_cost_sec += _time.perf_counter() - self._t_enter
         return None
 
     def __exit__(self, *exc):
+        t_exit = _time.perf_counter()
         close_event(self._event)
         from traceml_amd.core import arming as _arming
 
@@ -117,6 +127,9 @@ class _TraceStep:
         flush_step_events(step)
         self._recording.mark_trace_step_flushed()
         _kick_rank_stats(step)
+        global _self_cost_sec, _self_steps
+        _self_cost_sec += _time.perf_counter() - t_exit
+        _self_steps += 1
         return False
 
 
@@ -186,3 +199,11 @@ def trace_time(name: str):
         return wrapper
 
     return decorator
+
+
+def self_overhead_us_per_step():
+    """Mean microseconds the trace_step bracket itself spent per step
+    (enter+exit bookkeeping; excludes user code and async GPU stamps)."""
+    if _self_steps == 0:
+        return None
+    return _self_cost_sec / _self_steps * 1e6
