@@ -104,9 +104,14 @@ class _TraceStep:
         flags.optimizer_enabled = config.mode == "auto"
         if config.auto_optimizer_hooks:
             ensure_optimizer_timing_installed()
-        self._event = open_event(event_names.STEP_TIME)
+        # account the bookkeeping BEFORE the stamp launch: under deep
+        # pipelining a kernel launch can block on queue backpressure for a
+        # full step's worth of time — that is the stream's natural
+        # submission throttle (paid somewhere in any run), not our cost.
+        # Device-side stamp cost is measured separately (bench + rocprof).
         global _self_cost_sec
         _self_cost_sec += _time.perf_counter() - self._t_enter
+        self._event = open_event(event_names.STEP_TIME)
         return None
 
     def __exit__(self, *exc):
