@@ -111,3 +111,37 @@ def test_summary_pipeline_query_topology(db_path, monkeypatch):
     assert len(selects) == 2, selects
     assert "step_time_samples" in selects[0]
     assert "runtime_environment" in selects[1]
+
+
+def test_live_view_includes_issues_comm_stdout(db_path):
+    import json as _json
+    import sqlite3
+    import time as _time
+
+    scenarios.input_straggler(steps=30).write(db_path)
+    conn = sqlite3.connect(db_path)
+    with conn:
+        conn.execute(
+            "INSERT INTO rank_stats (global_rank, timestamp, "
+            "world_size_gathered, ranks_json) VALUES (0, ?, 4, ?)",
+            (_time.time(), _json.dumps([
+                {"rank": r, "step": 30, "input_ms": 4.0, "forward_ms": 30.0,
+                 "backward_ms": 235.0, "optimizer_ms": 8.0, "step_ms": 277.0,
+                 "ddp_comm_ms": 185.0, "peak_alloc_bytes": 0.0}
+                for r in range(4)
+            ])),
+        )
+        conn.execute(
+            "INSERT INTO stdout_stderr (global_rank, timestamp, stream, line)"
+            " VALUES (0, ?, 'stdout', 'epoch 3 loss 0.12')",
+            (_time.time(),),
+        )
+    conn.close()
+
+    payload = live_view(db_path)
+    kinds = [i["kind"] for i in payload["issues"]]
+    assert "INPUT_STRAGGLER" in kinds
+    assert payload["issues"][0]["severity"] == "crit"  # sorted
+    assert {i["section"] for i in payload["issues"]} >= {"step_time"}
+    assert len(payload["comm"]["ranks"]) == 4
+    assert payload["stdout"][-1]["line"] == "epoch 3 loss 0.12"

@@ -68,10 +68,31 @@ async function tick(){
       html+=`<div class="dim">${d.step_time.steps_analyzed} aligned steps · ${d.step_time.clock} clock · ${d.step_time.strategy}</div>`;
     }
     if(d.memory&&Object.keys(d.memory).length){
-      html+='<h2>Peak memory (GiB)</h2><table><tr><th>rank</th><th>allocated</th><th>reserved</th></tr>';
-      for(const [r,m] of Object.entries(d.memory))
-        html+=`<tr><td>r${r}</td><td>${m.alloc==null?'—':(m.alloc/2**30).toFixed(1)}</td><td>${m.reserved==null?'—':(m.reserved/2**30).toFixed(1)}</td></tr>`;
+      html+='<h2>Peak memory</h2><table><tr><th>rank</th><th>allocated</th><th>reserved</th><th style="text-align:left">of capacity</th></tr>';
+      for(const [r,m] of Object.entries(d.memory)){
+        const cap=m.capacity||288*2**30;
+        const pct=m.reserved?Math.min(100,m.reserved/cap*100):0;
+        const barColor=pct>92?'#d9534f':(pct>80?'#f0ad4e':'#1b998b');
+        html+=`<tr><td>r${r}</td><td>${m.alloc==null?'—':(m.alloc/2**30).toFixed(1)+' GiB'}</td><td>${m.reserved==null?'—':(m.reserved/2**30).toFixed(1)+' GiB'}</td>`+
+          `<td style="text-align:left;min-width:180px"><div class="bar" style="height:12px;background:#222"><div style="width:${pct}%;background:${barColor}"></div></div><span class="dim">${pct.toFixed(0)}% of ${(cap/2**30).toFixed(0)} GiB</span></td></tr>`;}
       html+='</table>';}
+    if(d.comm&&d.comm.ranks&&d.comm.ranks.length>1){
+      html+='<h2>RCCL rank stats (xGMI all-gather)</h2><table><tr><th>rank</th><th>step</th><th>input ms</th><th>bwd ms</th><th>ddp comm ms</th></tr>';
+      for(const r of d.comm.ranks)
+        html+=`<tr><td>r${r.rank}</td><td>${r.step}</td><td>${r.input_ms.toFixed(1)}</td><td>${r.backward_ms.toFixed(1)}</td><td>${r.ddp_comm_ms.toFixed(1)}</td></tr>`;
+      html+='</table>';}
+    if(d.issues&&d.issues.length){
+      html+='<h2>Findings</h2>';
+      for(const i of d.issues.slice(0,8)){
+        if(i.kind==='NORMAL'||i.kind==='BALANCED'||i.kind==='NO_DATA'||i.kind==='NO_GPU') continue;
+        html+=`<div class="verdict ${i.severity}" style="font-weight:400;padding:.4rem .8rem;margin:.3rem 0">`+
+          `<b>${i.status}</b> <span class="dim">[${i.section}]</span> ${i.summary}</div>`;}
+    }
+    if(d.stdout&&d.stdout.length){
+      html+='<h2>Console (rank 0)</h2><div style="background:#000;border-radius:6px;padding:.6rem;font-family:monospace;font-size:.75rem">';
+      for(const l of d.stdout)
+        html+=`<div style="color:${l.stream==='stderr'?'#e08':'#9d9'}">${l.line.replace(/</g,'&lt;')}</div>`;
+      html+='</div>';}
     if(d.system&&d.system.gpus&&Object.keys(d.system.gpus).length){
       html+='<h2>GPUs (amdsmi)</h2><table><tr><th>gpu</th><th>util %</th><th>VRAM GiB</th><th>temp °C</th><th>power W</th></tr>';
       for(const [g,m] of Object.entries(d.system.gpus))

@@ -65,6 +65,74 @@ def system_view(db_path: str) -> dict:
     }
 
 
+def issues_view(db_path: str, step_time_diagnosis) -> list:
+    """All sections' current findings, one flat severity-sorted list."""
+    from traceml_amd.diagnostics.common import SEVERITY_ORDER
+    from traceml_amd.diagnostics.process.api import (
+        diagnose_process,
+        load_process_context,
+    )
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+    from traceml_amd.diagnostics.system.api import (
+        diagnose_system,
+        load_system_context,
+    )
+
+    issues = []
+    for section, result in (
+        ("step_time", step_time_diagnosis),
+        ("step_memory", diagnose_step_memory(load_memory_series(db_path))),
+        ("system", diagnose_system(load_system_context(db_path))),
+        ("process", diagnose_process(load_process_context(db_path))),
+    ):
+        for issue in result.issues:
+            payload = issue.to_payload()
+            payload["section"] = section
+            issues.append(payload)
+    issues.sort(key=lambda i: -SEVERITY_ORDER.get(i.get("severity"), 0))
+    return issues
+
+
+def stdout_tail_view(db_path: str, n: int = 12) -> list:
+    import sqlite3
+
+    try:
+        conn = sqlite3.connect(f"file:{db_path}?mode=ro", uri=True)
+        try:
+            rows = conn.execute(
+                "SELECT stream, line FROM stdout_stderr ORDER BY id DESC LIMIT ?",
+                (n,),
+            ).fetchall()
+        finally:
+            conn.close()
+        return [{"stream": s, "line": l} for s, l in reversed(rows)]
+    except sqlite3.Error:
+        return []
+
+
+def comm_view(db_path: str) -> dict:
+    """Latest RCCL rank-stats gather (xGMI latency + per-rank skew)."""
+    import json as _json
+    import sqlite3
+
+    try:
+        conn = sqlite3.connect(f"file:{db_path}?mode=ro", uri=True)
+        try:
+            row = conn.execute(
+                "SELECT ranks_json FROM rank_stats ORDER BY id DESC LIMIT 1"
+            ).fetchone()
+        finally:
+            conn.close()
+        if not row or not row[0]:
+            return {}
+        return {"ranks": _json.loads(row[0])}
+    except (sqlite3.Error, ValueError):
+        return {}
+
+
 def live_view(db_path: str, session=None) -> dict:
     """One payload for live surfaces; optionally freshness-bridged via a
     LiveStepTimeSession."""
@@ -80,5 +148,8 @@ def live_view(db_path: str, session=None) -> dict:
         "step_time": step_time_view(result.window, result.diagnosis),
         "memory": memory_view(db_path),
         "system": system_view(db_path),
+        "issues": issues_view(db_path, result.diagnosis),
+        "stdout": stdout_tail_view(db_path),
+        "comm": comm_view(db_path),
     }
     return payload
