@@ -95,6 +95,24 @@ class CLIDisplayDriver(DisplayDriver):
                 f"[dim]{window.steps_analyzed} aligned steps · "
                 f"{window.clock} clock · strategy {window.training_strategy}[/dim]"
             )
+        # cross-section findings (memory/system/process), top 3 actionable
+        try:
+            from traceml_amd.renderers.views import issues_view
+
+            neutral = {"NORMAL", "BALANCED", "NO_DATA", "NO_GPU", "WARMUP"}
+            findings = [
+                i
+                for i in issues_view(db_path, result.diagnosis)
+                if i["kind"] not in neutral and i["section"] != "step_time"
+            ][:3]
+            for issue in findings:
+                style = _SEVERITY_STYLE.get(issue["severity"], "cyan")
+                renderables.append(
+                    f"[{style}]{issue['status']}[/] "
+                    f"[dim]\\[{issue['section']}][/dim] {issue['summary']}"
+                )
+        except Exception:
+            pass
         return Panel(Group(*renderables), title="traceml-amd live", border_style="blue")
 
     def stop(self) -> None:
