@@ -266,3 +266,18 @@ def test_trend_never_primary_over_other_findings(tmp_path):
     assert "STEP_TIME_DEGRADING" in kinds
     assert result.diagnosis.primary.kind != "STEP_TIME_DEGRADING"
     assert result.diagnosis.primary.kind == "COMPUTE_BOUND"
+
+
+def test_cli_view_html(tmp_path, capsys):
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+    from traceml_amd.reporting.final import generate_summary
+
+    db = str(tmp_path / "t.sqlite")
+    scenarios.input_bound(steps=30).write(db)
+    generate_summary(db, str(tmp_path))
+    out_html = str(tmp_path / "re.html")
+    assert main(["view", str(tmp_path / "final_summary.json"),
+                 "--html", out_html]) == 0
+    html = open(out_html).read()
+    assert "INPUT" in html and "<svg" in html

@@ -81,6 +81,8 @@ def build_parser() -> argparse.ArgumentParser:
 
     view_p = sub.add_parser("view", help="re-print a saved final summary")
     view_p.add_argument("summary_json")
+    view_p.add_argument("--html", metavar="OUT", default=None,
+                        help="also render a self-contained HTML report")
 
     inspect_p = sub.add_parser("inspect", help="dump per-rank msgpack backups")
     inspect_p.add_argument("path")
@@ -122,7 +124,7 @@ def main(argv: Optional[List[str]] = None) -> int:
     if args.command == "compare":
         return commands.run_compare(args.baseline, args.candidate)
     if args.command == "view":
-        return commands.run_view(args.summary_json)
+        return commands.run_view(args.summary_json, html_out=args.html)
     if args.command == "inspect":
         return commands.run_inspect(args.path)
     if args.command == "export-trace":

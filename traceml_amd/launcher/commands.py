@@ -237,8 +237,9 @@ def run_inspect(path: str) -> int:
     return 0
 
 
-def run_view(path: str) -> int:
-    """Re-print a saved final summary (reference: reporting/view/command.py:41)."""
+def run_view(path: str, html_out: Optional[str] = None) -> int:
+    """Re-print a saved final summary (reference: reporting/view/command.py:41);
+    optionally re-render the self-contained HTML report from the JSON."""
     import json
 
     try:
@@ -253,6 +254,12 @@ def run_view(path: str) -> int:
 
         text = build_verdict_text(payload)
     print(text)
+    if html_out:
+        from traceml_amd.reporting.html.document import render_html
+        from traceml_amd.utils.atomic_io import atomic_write_text
+
+        atomic_write_text(html_out, render_html(payload))
+        print(f"wrote {html_out}")
     return 0
 
 
