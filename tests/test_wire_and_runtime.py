@@ -186,3 +186,46 @@ def test_yaml_precedence(tmp_path, monkeypatch):
     assert s.interval == 3.0  # env beats yaml
     s = resolve_config(cli_overrides={"interval": 1.0})
     assert s.interval == 1.0  # cli beats env
+
+
+def test_aggregator_settle_writes_warning_on_missing_ranks(tmp_path, monkeypatch):
+    """expected_ranks=2 but only rank 0 says rank_finished: finalize still
+    completes and finalization_warning.json records the gap."""
+    import json
+
+    from tests import scenarios  # noqa: F401  (schema helpers)
+    from traceml_amd.aggregator.aggregator import TraceMLAggregator
+    from traceml_amd.runtime.settings import TraceMLSettings
+    from traceml_amd.telemetry.control import build_rank_finished
+    from traceml_amd.telemetry.envelope import build_telemetry_envelope
+    from traceml_amd.transport.tcp import TCPClient
+
+    monkeypatch.setenv("TRACEML_SESSION_ID", "settle")
+    settings = TraceMLSettings(
+        logs_dir=str(tmp_path), session_id="settle", aggregator_port=0,
+        finalize_timeout=6.0, expected_ranks=2, mode="cli",
+    )
+    aggregator = TraceMLAggregator(settings)
+    aggregator.start()
+    client = TCPClient("127.0.0.1", aggregator.port)
+    client.send_batch(
+        [
+            build_telemetry_envelope(
+                {"global_rank": 0, "pid": 1}, "step_time",
+                {"step_time_samples": [{"timestamp": 1.0, "step": 1,
+                                        "events": {}}]},
+            ),
+            build_rank_finished({"global_rank": 0, "pid": 1}),
+        ]
+    )
+    import time
+
+    time.sleep(1.0)
+    aggregator.stop()  # rank 1 never reports
+    client.close()
+    warning_path = tmp_path / "settle" / "finalization_warning.json"
+    assert warning_path.exists()
+    warning = json.loads(warning_path.read_text())
+    assert warning["expected_ranks"] == 2
+    assert warning["finished_ranks"] == [0]
+    assert (tmp_path / "settle" / "final_summary.json").exists()
