@@ -76,6 +76,26 @@ async function tick(){
         html+=`<tr><td>r${r}</td><td>${m.alloc==null?'—':(m.alloc/2**30).toFixed(1)+' GiB'}</td><td>${m.reserved==null?'—':(m.reserved/2**30).toFixed(1)+' GiB'}</td>`+
           `<td style="text-align:left;min-width:180px"><div class="bar" style="height:12px;background:#222"><div style="width:${pct}%;background:${barColor}"></div></div><span class="dim">${pct.toFixed(0)}% of ${(cap/2**30).toFixed(0)} GiB</span></td></tr>`;}
       html+='</table>';}
+    if(d.history&&Object.keys(d.history).length){
+      const ranksH=Object.keys(d.history);
+      let allPts=[];for(const r of ranksH)allPts=allPts.concat(d.history[r]);
+      if(allPts.length>4){
+        const xs=allPts.map(p=>p[0]),ys=allPts.map(p=>p[1]);
+        const x0=Math.min(...xs),x1=Math.max(...xs),y1=Math.max(...ys)*1.05||1;
+        const W=920,H=110;
+        const colors=['#7ab8ff','#1b998b','#e07b39','#d05ce3','#97cc04','#f0ad4e','#e85d75','#9ad'];
+        let svg=`<svg width="${W}" height="${H+18}" xmlns="http://www.w3.org/2000/svg">`;
+        svg+=`<line x1="0" y1="${H}" x2="${W}" y2="${H}" stroke="#333"/>`;
+        ranksH.forEach((r,i)=>{
+          const pts=d.history[r].map(p=>
+            `${((p[0]-x0)/Math.max(1,x1-x0)*W).toFixed(1)},${(H-p[1]/y1*H).toFixed(1)}`).join(' ');
+          svg+=`<polyline points="${pts}" fill="none" stroke="${colors[i%8]}" stroke-width="1.5"/>`;});
+        svg+=`<text x="4" y="12" font-size="10" fill="#888">step time ms (0–${y1.toFixed(1)}), steps ${x0}–${x1}</text>`;
+        svg+=ranksH.map((r,i)=>`<text x="${70+i*46}" y="${H+14}" font-size="10" fill="${colors[i%8]}">r${r}</text>`).join('');
+        svg+='</svg>';
+        html+='<h2>Step-time history</h2>'+svg;
+      }
+    }
     if(d.comm&&d.comm.ranks&&d.comm.ranks.length>1){
       html+='<h2>RCCL rank stats (xGMI all-gather)</h2><table><tr><th>rank</th><th>step</th><th>input ms</th><th>bwd ms</th><th>ddp comm ms</th></tr>';
       for(const r of d.comm.ranks)

@@ -133,6 +133,14 @@ def comm_view(db_path: str) -> dict:
         return {}
 
 
+def history_view(window, max_points: int = 120) -> dict:
+    """Per-rank step-time series for charts: {rank: [[step, ms], ...]}."""
+    out = {}
+    for rank, series in window.step_series.items():
+        out[str(rank)] = [[s, round(ms, 3)] for s, ms in series[-max_points:]]
+    return out
+
+
 def live_view(db_path: str, session=None) -> dict:
     """One payload for live surfaces; optionally freshness-bridged via a
     LiveStepTimeSession."""
@@ -151,5 +159,6 @@ def live_view(db_path: str, session=None) -> dict:
         "issues": issues_view(db_path, result.diagnosis),
         "stdout": stdout_tail_view(db_path),
         "comm": comm_view(db_path),
+        "history": history_view(result.window),
     }
     return payload
