@@ -105,6 +105,10 @@ class DeepProfile:
                 module.register_full_backward_hook(self._backward_post(bwd_name))
             )
 
+    # NOTE: with backward=True, torch emits a one-time warning for the first
+    # layer when the model inputs don't require grad ("Full backward hook is
+    # firing when gradients are computed with respect to module outputs") —
+    # harmless: the bracket still measures that module's grad computation.
     def _backward_pre(self, name: str):
         pre = self._pre(name)
 

@@ -70,6 +70,11 @@ class TCPServer:
                 daemon=True,
             )
             t.start()
+            # prune finished reader threads so long runs with reconnecting
+            # clients don't accumulate dead references
+            self._client_threads = [
+                th for th in self._client_threads if th.is_alive()
+            ]
             self._client_threads.append(t)
 
     def _handle_client(self, client: socket.socket) -> None:
