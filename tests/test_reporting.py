@@ -139,3 +139,41 @@ def test_no_manifest_no_hint(db_path, tmp_path):
     scenarios.input_bound(steps=30).write(db_path)
     payload = generate_summary(db_path, str(tmp_path))
     assert "num_workers=" not in payload["primary_diagnosis"]["action"]
+
+
+def test_step_memory_common_step_alignment(db_path, tmp_path):
+    """A rank whose memory spiked AFTER the common window ended must not
+    contribute that spike to the aligned comparison."""
+    import sqlite3
+    import time as _time
+
+    from traceml_amd.aggregator.writers import build_all_writers
+    from traceml_amd.reporting.sections.step_memory import build as build_mem
+
+    gib = 1 << 30
+    conn = sqlite3.connect(db_path)
+    for w in build_all_writers():
+        w.init_schema(conn)
+    with conn:
+        for rank in (0, 1):
+            last = 30 if rank == 0 else 20  # rank 1 died at step 20
+            for step in range(1, last + 1):
+                alloc = 10 * gib
+                if rank == 0 and step > 20:
+                    alloc = 200 * gib  # spike outside the common window
+                conn.execute(
+                    "INSERT INTO step_memory_samples (global_rank, world_size,"
+                    " timestamp, step, peak_allocated_bytes,"
+                    " peak_reserved_bytes, device_capacity_bytes, device)"
+                    " VALUES (?, 2, ?, ?, ?, ?, ?, 'cuda:0')",
+                    (rank, _time.time() + step, step, alloc, alloc + gib,
+                     288 * gib),
+                )
+    conn.close()
+    payload = build_mem(db_path)
+    rows = payload["groups"]["rows"]
+    # aligned window = steps 1..20 for both ranks: the spike is excluded
+    assert rows["0"]["metrics"]["peak_allocated_bytes"] == 10 * gib
+    assert rows["1"]["metrics"]["peak_allocated_bytes"] == 10 * gib
+    assert payload["global"]["window"]["steps_analyzed"] == 20
+    assert payload["global"]["window"]["end_step"] == 20

@@ -18,15 +18,35 @@ def build(db_path: str) -> dict:
     series = load_memory_series(db_path)
     payload.update(diagnose_step_memory(series).to_payload())
 
+    # common-step alignment (the section's declared contract): compare peak
+    # watermarks over the steps EVERY rank reported, so a rank that died
+    # early does not skew the comparison window
+    common_steps = None
+    for s in series.values():
+        steps = set(s.steps)
+        common_steps = steps if common_steps is None else (common_steps & steps)
+    common_steps = common_steps or set()
+
     per_rank = {}
     latest_step = None
+    window_start = window_end = None
     for rank, s in sorted(series.items()):
-        alloc = [v for v in s.peak_allocated if v is not None]
-        reserved = [v for v in s.peak_reserved if v is not None]
+        pairs = [
+            (step, alloc, reserved)
+            for step, alloc, reserved in zip(
+                s.steps, s.peak_allocated, s.peak_reserved
+            )
+            if not common_steps or step in common_steps
+        ]
+        alloc = [a for _, a, _ in pairs if a is not None]
+        reserved = [r for _, _, r in pairs if r is not None]
         per_rank[str(rank)] = {
             "peak_allocated_bytes": max(alloc) if alloc else None,
             "peak_reserved_bytes": max(reserved) if reserved else None,
         }
+        if pairs:
+            window_start = pairs[0][0] if window_start is None else min(window_start, pairs[0][0])
+            window_end = pairs[-1][0] if window_end is None else max(window_end, pairs[-1][0])
         if s.steps:
             latest_step = max(latest_step or 0, s.steps[-1])
 
@@ -40,6 +60,11 @@ def build(db_path: str) -> dict:
 
     payload["global"]["window"]["kind"] = "step_window"
     payload["global"]["window"]["alignment"] = "common_steps"
+    payload["global"]["window"]["start_step"] = window_start
+    payload["global"]["window"]["end_step"] = window_end
+    payload["global"]["window"]["steps_analyzed"] = (
+        len(common_steps) or None
+    )
     fill_metric_maps(payload, STEP_MEMORY_METRICS, per_rank)
     for rank, s in series.items():
         row = payload["groups"]["rows"].get(str(rank))
