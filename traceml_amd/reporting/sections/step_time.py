@@ -65,8 +65,40 @@ def build(db_path: str) -> dict:
         "cohorts": window.cohorts,
         "training_strategy": window.training_strategy,
     }
+    comm = _latest_rank_stats(db_path)
+    if comm:
+        payload["evidence_extra"]["rccl_rank_stats"] = comm
     payload["card"] = _card(window, payload)
     return payload
+
+
+def _latest_rank_stats(db_path: str) -> dict:
+    """Most recent RCCL-over-xGMI gather (per-rank skew + gather latency)."""
+    import json as _json
+    import sqlite3
+
+    try:
+        conn = sqlite3.connect(f"file:{db_path}?mode=ro", uri=True)
+        try:
+            row = conn.execute(
+                "SELECT timestamp, world_size_gathered, ranks_json "
+                "FROM rank_stats ORDER BY id DESC LIMIT 1"
+            ).fetchone()
+        finally:
+            conn.close()
+    except sqlite3.Error:
+        return {}
+    if not row or not row[2]:
+        return {}
+    try:
+        ranks = _json.loads(row[2])
+    except ValueError:
+        return {}
+    return {
+        "timestamp": row[0],
+        "world_size": row[1],
+        "ranks": ranks,
+    }
 
 
 def _card(window, payload) -> str:
