@@ -1,0 +1,96 @@
+"""Unit coverage for the smaller utilities: codec fallback, atomic io,
+bands, manifest transitions, yaml loader edges."""
+
+import json
+import os
+
+import pytest
+
+
+def test_codec_roundtrip_and_default():
+    import numpy as np
+
+    from traceml_amd.transport import codec
+
+    payload = {"a": 1, "b": [1.5, "x"], "c": {"n": None}}
+    assert codec.decode(codec.encode(payload)) == payload
+    # numpy scalars survive via the default hook
+    out = codec.decode(codec.encode({"v": np.float32(2.5)}))
+    assert out["v"] == pytest.approx(2.5)
+
+
+def test_atomic_write_replaces_not_partial(tmp_path):
+    from traceml_amd.utils.atomic_io import atomic_write_json
+
+    path = str(tmp_path / "x.json")
+    atomic_write_json(path, {"v": 1})
+    atomic_write_json(path, {"v": 2})
+    assert json.load(open(path)) == {"v": 2}
+    assert [p for p in os.listdir(tmp_path) if p.startswith(".tmp-")] == []
+
+
+def test_bands_classify():
+    from traceml_amd.diagnostics.bands import BandThresholds
+
+    band = BandThresholds(warn=0.8, crit=0.9)
+    assert band.classify(None) is None
+    assert band.classify(0.5) is None
+    assert band.classify(0.85) == "warn"
+    assert band.classify(0.95) == "crit"
+
+
+def test_manifest_status_transitions(tmp_path):
+    from traceml_amd.launcher import manifest
+
+    sdir = str(tmp_path)
+    manifest.write_run_manifest(sdir, manifest.STATUS_STARTING, script="t.py")
+    manifest.update_status(sdir, manifest.STATUS_RUNNING)
+    manifest.update_status(
+        sdir, manifest.STATUS_COMPLETED, extra={"exit_code": 0}
+    )
+    data = json.load(open(manifest.manifest_path(sdir)))
+    assert data["status"] == "completed"
+    assert data["exit_code"] == 0
+    assert data["script"] == "t.py"  # earlier fields preserved
+
+
+def test_yaml_loader_ignores_unknown_and_bad_yaml(tmp_path, monkeypatch):
+    from traceml_amd.config.yaml_loader import load_yaml_settings
+
+    good = tmp_path / "traceml.yaml"
+    good.write_text("interval: 1.5\nnot_a_setting: 7\nmode: cli\n")
+    loaded = load_yaml_settings(str(good))
+    assert loaded == {"interval": 1.5, "mode": "cli"}
+
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("{{{{not yaml")
+    assert load_yaml_settings(str(bad)) == {}
+
+
+def test_yaml_walkup(tmp_path, monkeypatch):
+    from traceml_amd.config.yaml_loader import find_yaml
+
+    (tmp_path / "traceml.yaml").write_text("interval: 2.0\n")
+    nested = tmp_path / "a" / "b" / "c"
+    nested.mkdir(parents=True)
+    found = find_yaml(str(nested))
+    assert found == str(tmp_path / "traceml.yaml")
+
+
+def test_read_msgpack_table_truncated(tmp_path):
+    """A torn final record (crash mid-write) must not break inspect."""
+    from traceml_amd.database.database import Database
+    from traceml_amd.database.writer import DatabaseWriter, read_msgpack_table
+
+    db = Database()
+    db.add_record("t", {"v": 1})
+    db.add_record("t", {"v": 2})
+    writer = DatabaseWriter("s", db, str(tmp_path))
+    writer.flush()
+    writer.close()
+    path = str(tmp_path / "s" / "t.msgpack")
+    blob = open(path, "rb").read()
+    open(path, "wb").write(blob[:-3])  # tear the tail
+    rows = read_msgpack_table(path)
+    assert rows[0] == {"v": 1}
+    assert len(rows) == 1  # torn record dropped, no exception
