@@ -281,3 +281,15 @@ def test_cli_view_html(tmp_path, capsys):
                  "--html", out_html]) == 0
     html = open(out_html).read()
     assert "INPUT" in html and "<svg" in html
+
+
+def test_cli_top_snapshot(tmp_path, capsys):
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+
+    db = str(tmp_path / "telemetry.sqlite")
+    scenarios.input_straggler(steps=30).write(db)
+    assert main(["top", db]) == 0
+    out = capsys.readouterr().out
+    assert "INPUT STRAGGLER" in out
+    assert "r2" in out and "aligned steps" in out

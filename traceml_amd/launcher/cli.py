@@ -87,6 +87,11 @@ def build_parser() -> argparse.ArgumentParser:
     inspect_p = sub.add_parser("inspect", help="dump per-rank msgpack backups")
     inspect_p.add_argument("path")
 
+    top_p = sub.add_parser(
+        "top", help="one-shot snapshot of a (possibly live) session"
+    )
+    top_p.add_argument("session", help="session dir or telemetry.sqlite path")
+
     trace_p = sub.add_parser(
         "export-trace",
         help="export the step-time history as a chrome://tracing timeline",
@@ -127,6 +132,8 @@ def main(argv: Optional[List[str]] = None) -> int:
         return commands.run_view(args.summary_json, html_out=args.html)
     if args.command == "inspect":
         return commands.run_inspect(args.path)
+    if args.command == "top":
+        return commands.run_top(args.session)
     if args.command == "export-trace":
         from traceml_amd.reporting.trace_export import export_chrome_trace
 

@@ -267,3 +267,45 @@ def run_compare(path_a: str, path_b: str) -> int:
     from traceml_amd.reporting.compare.command import compare_files
 
     return compare_files(path_a, path_b)
+
+
+def run_top(session: str) -> int:
+    """One-shot live snapshot: verdict, per-rank phase table, findings."""
+    db_path = session
+    if os.path.isdir(session):
+        from traceml_amd.sdk import protocol
+
+        db_path = protocol.sqlite_path(session)
+        if not os.path.exists(db_path):
+            # maybe they passed the logs dir's session child already
+            alt = os.path.join(session, "aggregator", "telemetry.sqlite")
+            db_path = alt if os.path.exists(alt) else db_path
+    if not os.path.exists(db_path):
+        print(f"no telemetry db at {db_path}", file=sys.stderr)
+        return 1
+    from traceml_amd.renderers import live_view
+
+    payload = live_view(db_path)
+    st = payload["step_time"]
+    diag = st["diagnosis"]
+    print(f"{diag.get('status')} [{diag.get('severity')}] — {diag.get('summary')}")
+    ranks = st.get("ranks", {})
+    if ranks:
+        metrics = [m for m in st["table_metrics"]
+                   if any(ranks[r].get(m) is not None for r in ranks)]
+        header = "rank    " + "".join(f"{m.replace('_ms',''):>12}" for m in metrics)
+        print(header)
+        for r in sorted(ranks, key=lambda k: (len(k), k)):
+            row = f"r{r:<6}"
+            for m in metrics:
+                v = ranks[r].get(m)
+                row += f"{v:>12.1f}" if v is not None else f"{'—':>12}"
+            print(row)
+        print(f"({st['steps_analyzed']} aligned steps, {st['clock']} clock, "
+              f"{payload['freshness']})")
+    neutral = {"NORMAL", "BALANCED", "NO_DATA", "NO_GPU", "WARMUP"}
+    for issue in payload.get("issues", []):
+        if issue["kind"] in neutral:
+            continue
+        print(f"  {issue['status']} [{issue['section']}] {issue['summary']}")
+    return 0
