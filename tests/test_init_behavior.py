@@ -72,3 +72,25 @@ def test_custom_mode_selective_patches():
     assert config.patch_forward is False
     assert config.patch_backward is True
     assert config.patch_h2d is False
+
+
+def test_lazy_api_dir_and_alias():
+    import traceml_amd
+
+    exported = dir(traceml_amd)
+    for sym in ("init", "trace_step", "summary", "final_summary",
+                "wrap_forward", "deep_profile"):
+        assert sym in exported
+
+    import warnings
+
+    with warnings.catch_warnings(record=True) as caught:
+        warnings.simplefilter("always")
+        import importlib
+
+        import traceml
+
+        importlib.reload(traceml)
+    assert any(issubclass(w.category, DeprecationWarning) for w in caught)
+    assert traceml.__version__ == traceml_amd.__version__
+    assert traceml.trace_step is not None
