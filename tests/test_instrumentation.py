@@ -173,3 +173,43 @@ def test_dataloader_timing_with_worker_processes(armed_auto_config, tiny_model):
     for row in rows:
         assert event_names.DATALOADER in row["events"]
         assert row["events"][event_names.DATALOADER]["cpu_ms"] >= 0.0
+
+
+def test_bracket_overhead_regression_guard(armed_auto_config):
+    """The per-step bracket must stay in the ~0.1 ms range on CPU. Guard
+    threshold is generous (1.5 ms) to stay robust on slow CI boxes while
+    still catching accidental O(model)/O(history) regressions."""
+    import time
+
+    model = nn.Linear(8, 8)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+
+    def step(traced):
+        if traced:
+            with trace_step(model):
+                opt.zero_grad()
+                model(torch.randn(4, 8)).sum().backward()
+                opt.step()
+        else:
+            opt.zero_grad()
+            model(torch.randn(4, 8)).sum().backward()
+            opt.step()
+
+    from traceml_amd.core import timing
+
+    for _ in range(100):
+        step(True)
+        step(False)
+    n = 500
+    timing.clear_for_tests()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        step(False)
+    base = time.perf_counter() - t0
+    timing.clear_for_tests()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        step(True)
+    traced = time.perf_counter() - t0
+    overhead_us = (traced - base) / n * 1e6
+    assert overhead_us < 1500, f"bracket overhead regressed: {overhead_us:.0f}us"
