@@ -293,3 +293,19 @@ def test_cli_top_snapshot(tmp_path, capsys):
     out = capsys.readouterr().out
     assert "INPUT STRAGGLER" in out
     assert "r2" in out and "aligned steps" in out
+
+
+def test_cli_top_accepts_session_dir(tmp_path, capsys):
+    import os
+
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+
+    session = tmp_path / "sess"
+    os.makedirs(session / "aggregator")
+    scenarios.healthy_ddp(ranks=2, steps=25).write(
+        str(session / "aggregator" / "telemetry.sqlite")
+    )
+    assert main(["top", str(session)]) == 0
+    out = capsys.readouterr().out
+    assert "aligned steps" in out and "r0" in out and "r1" in out

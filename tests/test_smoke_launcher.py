@@ -253,3 +253,33 @@ def test_trace_max_steps_budget_e2e(tmp_path):
         (tmp_path / "logs" / "budget" / "final_summary.json").read_text()
     )
     assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 25
+
+
+@pytest.mark.timeout(400)
+def test_compute_straggler_demo_e2e(tmp_path):
+    """Second flagship scenario through the full stack: rank 1 burns extra
+    forward compute -> COMPUTE_STRAGGLER with culprit r1."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "40"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--nproc-per-node", "4",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "cstrag",
+            "--aggregator-port", "29887",
+            "--master-port", "29607",
+            os.path.join(REPO_ROOT, "examples", "demo",
+                         "mlp_ddp_compute_straggler.py"),
+        ],
+        env=env, capture_output=True, text=True, timeout=380, cwd=REPO_ROOT,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    payload = json.loads(
+        (tmp_path / "logs" / "cstrag" / "final_summary.json").read_text()
+    )
+    primary = payload["primary_diagnosis"]
+    assert primary["kind"] in ("COMPUTE_STRAGGLER", "STRAGGLER"), primary["kind"]
+    assert payload["step_time"]["diagnosis"]["ranks"] == [1]
