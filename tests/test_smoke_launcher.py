@@ -225,6 +225,10 @@ def test_four_rank_input_straggler_e2e(tmp_path):
     assert st_diag["ranks"] == [2]
     assert "ddp_comm_ms_per_rank" in st_diag["evidence"]
     assert payload["step_time"]["metadata"]["global_ranks_used"] == [0, 1, 2, 3]
+    # the executor's lazily-armed RCCL(gloo) rank-stats exchange flowed
+    # into the final summary's evidence
+    comm = payload["step_time"]["evidence_extra"].get("rccl_rank_stats")
+    assert comm and len(comm["ranks"]) == 4, comm
 
 
 @pytest.mark.timeout(240)
