@@ -168,3 +168,13 @@ def test_cohorts_partition_ranks():
     assert set(window.cohorts["slow"]) == {2}
     assert set(window.cohorts["fast"]) == {3}
     assert set(window.cohorts["typical"]) == {0, 1}
+
+
+def test_window_size_cap_takes_trailing_steps():
+    """The analyzer honors window_size by keeping the TRAILING steps."""
+    rows = [_row(0, s, _events(fwd=float(s))) for s in range(1, 31)]
+    window = StepTimeAnalyzer(window_size=10).analyze(rows)
+    assert window.steps_analyzed == 10
+    assert window.start_step == 21 and window.end_step == 30
+    # mean of fwd over steps 21..30
+    assert window.ranks[0].forward_ms == pytest.approx(25.5)
