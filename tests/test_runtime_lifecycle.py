@@ -309,3 +309,25 @@ def test_cli_top_accepts_session_dir(tmp_path, capsys):
     assert main(["top", str(session)]) == 0
     out = capsys.readouterr().out
     assert "aligned steps" in out and "r0" in out and "r1" in out
+
+
+def test_compare_fail_on_regression_gate(tmp_path, capsys):
+    from tests import scenarios
+    from traceml_amd.launcher.cli import main
+    from traceml_amd.reporting.final import generate_summary
+
+    fast_db = str(tmp_path / "fast.sqlite")
+    slow_db = str(tmp_path / "slow.sqlite")
+    scenarios.healthy_ddp(ranks=1, steps=30).write(fast_db)
+    scenarios.input_bound(steps=30).write(slow_db)
+    fast_dir = tmp_path / "fast"
+    slow_dir = tmp_path / "slow"
+    generate_summary(fast_db, str(fast_dir))
+    generate_summary(slow_db, str(slow_dir))
+    fast = str(fast_dir / "final_summary.json")
+    slow = str(slow_dir / "final_summary.json")
+
+    assert main(["compare", fast, slow]) == 0  # default: report only
+    capsys.readouterr()
+    assert main(["compare", fast, slow, "--fail-on-regression"]) == 4
+    assert main(["compare", slow, fast, "--fail-on-regression"]) == 0

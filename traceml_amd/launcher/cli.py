@@ -78,6 +78,10 @@ def build_parser() -> argparse.ArgumentParser:
     compare_p = sub.add_parser("compare", help="compare two final summaries")
     compare_p.add_argument("baseline")
     compare_p.add_argument("candidate")
+    compare_p.add_argument(
+        "--fail-on-regression", action="store_true",
+        help="exit 4 when the verdict is REGRESSION (CI perf gate)",
+    )
 
     view_p = sub.add_parser("view", help="re-print a saved final summary")
     view_p.add_argument("summary_json")
@@ -127,7 +131,10 @@ def main(argv: Optional[List[str]] = None) -> int:
     if args.command == "serve":
         return commands.run_serve(_overrides(args))
     if args.command == "compare":
-        return commands.run_compare(args.baseline, args.candidate)
+        return commands.run_compare(
+            args.baseline, args.candidate,
+            fail_on_regression=args.fail_on_regression,
+        )
     if args.command == "view":
         return commands.run_view(args.summary_json, html_out=args.html)
     if args.command == "inspect":

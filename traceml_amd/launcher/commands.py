@@ -263,10 +263,14 @@ def run_view(path: str, html_out: Optional[str] = None) -> int:
     return 0
 
 
-def run_compare(path_a: str, path_b: str) -> int:
+def run_compare(
+    path_a: str, path_b: str, fail_on_regression: bool = False
+) -> int:
     from traceml_amd.reporting.compare.command import compare_files
 
-    return compare_files(path_a, path_b)
+    return compare_files(
+        path_a, path_b, fail_on_regression=fail_on_regression
+    )
 
 
 def run_top(session: str) -> int:

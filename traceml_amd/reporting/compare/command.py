@@ -233,7 +233,9 @@ def render_compare(result: dict) -> str:
     return "\n".join(lines)
 
 
-def compare_files(path_a: str, path_b: str) -> int:
+def compare_files(
+    path_a: str, path_b: str, fail_on_regression: bool = False
+) -> int:
     try:
         baseline = load_summary(path_a)
         candidate = load_summary(path_b)
@@ -242,4 +244,6 @@ def compare_files(path_a: str, path_b: str) -> int:
         return 1
     result = compare_payloads(baseline, candidate)
     print(render_compare(result))
+    if fail_on_regression and result["verdict"] == "REGRESSION":
+        return 4  # CI perf gate
     return 0
