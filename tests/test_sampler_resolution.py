@@ -137,3 +137,21 @@ def test_publisher_one_payload_per_sampler_per_tick():
     assert {p["meta"]["sampler"] for p in sent[0]} == {"a", "b"}
     publisher.publish()
     assert len(sent) == 1  # nothing new -> nothing sent
+
+
+def test_step_time_queue_drop_oldest(monkeypatch):
+    """Handoff queue at capacity drops the OLDEST batch (bounded memory;
+    live telemetry prefers fresh data)."""
+    monkeypatch.setattr(timing, "STEP_TIME_QUEUE_MAX", 5)
+    timing.clear_for_tests()
+    for step in range(1, 11):
+        timing.record_event(
+            timing.TimeEvent(
+                name=event_names.FORWARD, device="cpu",
+                cpu_start=1.0, cpu_end=1.01,
+            )
+        )
+        timing.flush_step_time_buffer(step)
+    batches = timing.drain_step_time_queue()
+    assert [b.step for b in batches] == [6, 7, 8, 9, 10]
+    timing.clear_for_tests()
