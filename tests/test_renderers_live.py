@@ -145,3 +145,14 @@ def test_live_view_includes_issues_comm_stdout(db_path):
     assert {i["section"] for i in payload["issues"]} >= {"step_time"}
     assert len(payload["comm"]["ranks"]) == 4
     assert payload["stdout"][-1]["line"] == "epoch 3 loss 0.12"
+
+
+def test_history_view_caps_points(db_path):
+    from traceml_amd.renderers.views import history_view
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    scenarios.healthy_ddp(ranks=1, steps=200).write(db_path)
+    window = StepTimePipeline(db_path, profile="summary").run().window
+    history = history_view(window, max_points=50)
+    assert len(history["0"]) == 50
+    assert history["0"][-1][0] == 200  # newest step kept
