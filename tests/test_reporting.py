@@ -177,3 +177,25 @@ def test_step_memory_common_step_alignment(db_path, tmp_path):
     assert rows["1"]["metrics"]["peak_allocated_bytes"] == 10 * gib
     assert payload["global"]["window"]["steps_analyzed"] == 20
     assert payload["global"]["window"]["end_step"] == 20
+
+
+def test_html_renderer_edge_cases():
+    from traceml_amd.reporting.html.document import (
+        _phase_bar_svg,
+        _rank_table,
+        render_html,
+    )
+
+    assert _phase_bar_svg({}) == ""
+    assert _phase_bar_svg({"input": None, "compute": None}) == ""
+    assert _rank_table({"groups": {"rows": {}}, "metadata": {}}) == ""
+    # minimal payload renders without crashing
+    html = render_html(
+        {
+            "primary_diagnosis": {"status": "X", "summary": "s",
+                                  "severity": "info"},
+            "generated_at": "t",
+            "schema_version": 1.7,
+        }
+    )
+    assert "<html" in html and "X" in html
