@@ -331,3 +331,20 @@ def test_compare_fail_on_regression_gate(tmp_path, capsys):
     capsys.readouterr()
     assert main(["compare", fast, slow, "--fail-on-regression"]) == 4
     assert main(["compare", slow, fast, "--fail-on-regression"]) == 0
+
+
+def test_export_trace_max_steps(tmp_path):
+    from tests import scenarios
+    from traceml_amd.reporting.trace_export import build_chrome_trace
+
+    db = str(tmp_path / "t.sqlite")
+    scenarios.healthy_ddp(ranks=2, steps=50).write(db)
+    trace = build_chrome_trace(db, max_steps=5)
+    step_events = [
+        e for e in trace["traceEvents"]
+        if e.get("ph") == "X" and e["name"].startswith("step ")
+    ]
+    # 2 ranks x 5 trailing steps
+    assert len(step_events) == 10
+    steps = sorted({e["args"]["step"] for e in step_events})
+    assert steps == [46, 47, 48, 49, 50]
