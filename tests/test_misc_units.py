@@ -94,3 +94,25 @@ def test_read_msgpack_table_truncated(tmp_path):
     rows = read_msgpack_table(path)
     assert rows[0] == {"v": 1}
     assert len(rows) == 1  # torn record dropped, no exception
+
+
+def test_all_primary_promotable_kinds_have_actions():
+    """UX completeness: every promotable verdict the rules can emit carries
+    a non-empty action string."""
+    from traceml_amd.diagnostics.step_time.rules import _ACTIONS
+    from traceml_amd.reporting.primary import _PROMOTABLE
+
+    for kind in _PROMOTABLE:
+        assert _ACTIONS.get(kind), f"{kind} has no action text"
+
+
+def test_event_name_vocabulary_is_closed():
+    """The wire vocabulary, the analyzer's signal map and the summary-key
+    map agree (a renamed event would silently drop a phase)."""
+    from traceml_amd.core import event_names, timing
+    from traceml_amd.steptime.model import STEP_TIME_EVENT_NAMES
+
+    assert set(event_names.ALL_EVENT_NAMES) == set(STEP_TIME_EVENT_NAMES)
+    assert set(timing._SUMMARY_KEYS) <= set(event_names.ALL_EVENT_NAMES)
+    for name in event_names.ALL_EVENT_NAMES:
+        assert name.startswith(event_names.PREFIX)
