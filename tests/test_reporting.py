@@ -199,3 +199,25 @@ def test_html_renderer_edge_cases():
         }
     )
     assert "<html" in html and "X" in html
+
+
+def test_compare_render_includes_per_rank_lines(db_path, tmp_path):
+    from traceml_amd.reporting.compare.command import (
+        compare_payloads,
+        render_compare,
+    )
+
+    scenarios.healthy_ddp(ranks=4, steps=30).write(db_path)
+    a = FinalReportGenerator(db_path).generate()
+    slow_db = str(tmp_path / "slow.sqlite")
+    scenario = scenarios.healthy_ddp(ranks=4, steps=30)
+    for profile in scenario.profiles.values():
+        profile.backward_ms = 200.0  # uniformly slower candidate
+    scenario.write(slow_db)
+    b = FinalReportGenerator(slow_db).generate()
+
+    text = render_compare(compare_payloads(a, b))
+    assert "step_time_ms by rank:" in text
+    for rank in ("r0", "r1", "r2", "r3"):
+        assert rank in text
+    assert "REGRESSION" in text
