@@ -75,8 +75,14 @@ class SQLiteWriterSimple:
     def ingest(self, envelope: dict) -> None:
         with self._queue_lock:
             while len(self._queue) >= QUEUE_MAX:
-                self._queue.popleft()
-                self.dropped += 1
+                evicted = self._queue.popleft()
+                if isinstance(evicted, _FlushBarrier):
+                    # Never strand a force_flush() waiter on its full timeout:
+                    # release it immediately (durability is degraded under
+                    # overload anyway — rows ahead of it were just dropped).
+                    evicted.event.set()
+                else:
+                    self.dropped += 1
             self._queue.append(envelope)
         self._wake.set()
 

@@ -159,6 +159,48 @@ def get_backend():
         return _backend
 
 
+def get_backend_or_none():
+    """Fail-open variant for hot/sampler paths: an unavailable native timer
+    degrades to None (CPU clock) instead of raising. The loud check lives in
+    the init() preflight (sdk/initial.py)."""
+    try:
+        return get_backend()
+    except GpuTimerUnavailable:
+        return None
+
+
+def preflight_check() -> None:
+    """Verify the native extension is loadable when a GPU is visible.
+
+    Called once from ``init()`` so a GPU machine with a stale/missing ``.so``
+    refuses (or warns+no-ops, per the fail-open ladder) BEFORE training
+    starts — never at first mark inside the user's hot loop. Deliberately
+    does not bind a device: ranks may ``torch.cuda.set_device`` after
+    ``init()``, so full backend construction stays lazy at first mark.
+    """
+    mode = os.environ.get("TRACEML_AMD_GPU_TIMER", "native").lower()
+    if mode in ("off", "torch"):
+        return
+    try:
+        import torch
+
+        if not torch.cuda.is_available():
+            return
+    except Exception:
+        return
+    try:
+        from traceml_amd.ops import hip_ext
+
+        hip_ext.load_extension()
+    except Exception as exc:
+        raise GpuTimerUnavailable(
+            "traceml_amd: GPU present but the native _traceml_hip extension "
+            "could not be loaded. Build it with "
+            "`python -m traceml_amd.ops.build_ext` (gfx950). "
+            f"Underlying error: {exc!r}"
+        ) from exc
+
+
 def reset_backend_for_tests() -> None:
     global _backend, _backend_resolved
     with _lock:

@@ -35,6 +35,15 @@ class StepMemoryEvent:
 _queue_lock = threading.Lock()
 _queue: Deque[StepMemoryEvent] = deque()
 
+#: most recent per-step peak-allocated watermark (bytes); feeds the RCCL
+#: rank-stats gather so peers see each other's memory pressure. None until
+#: the first GPU-measured step (null ≠ 0).
+_last_peak_alloc: Optional[int] = None
+
+
+def last_peak_alloc_bytes() -> Optional[int]:
+    return _last_peak_alloc
+
 
 def _cuda():
     try:
@@ -89,6 +98,9 @@ class StepMemoryTracker:
             device_capacity_bytes=self._capacity,
             device=device,
         )
+        if peak_alloc is not None:
+            global _last_peak_alloc
+            _last_peak_alloc = peak_alloc
         with _queue_lock:
             if len(_queue) >= STEP_MEMORY_QUEUE_MAX:
                 _queue.popleft()
@@ -103,5 +115,7 @@ def drain_step_memory_queue() -> List[StepMemoryEvent]:
 
 
 def clear_for_tests() -> None:
+    global _last_peak_alloc
     with _queue_lock:
         _queue.clear()
+    _last_peak_alloc = None

@@ -62,7 +62,7 @@ class TimeEvent:
         """
         if self._gpu_done or not self.has_gpu:
             return True
-        backend = gpu_timer.get_backend()
+        backend = gpu_timer.get_backend_or_none()
         if backend is None:
             # GPU handles without a backend: device gone, drop the GPU side.
             self._gpu_done = True
@@ -97,13 +97,31 @@ _step_time_queue: Deque[StepTimeBatch] = deque()
 _dropped_batches = 0
 
 
+_warned_timer_unavailable = False
+
+
 def _gpu_mark_if_enabled(name: str) -> Optional[int]:
     if name in _CPU_ONLY_EVENTS:
         return None
-    # A GPU machine without the native extension must fail LOUDLY (policy:
-    # a silent CPU-clock fallback could masquerade as the native path), so
-    # GpuTimerUnavailable propagates; only transient mark errors degrade.
-    backend = gpu_timer.get_backend()
+    # Fail-open here: the loud check for a GPU machine with a missing native
+    # extension happens ONCE at init() preflight (sdk/initial.py), never
+    # inside the user's hot loop — telemetry must not crash training
+    # (reference principle: docs/developer_guide/architecture.md:54-59).
+    # The preflight is what keeps the native-path signal trustworthy; by the
+    # time a mark runs, an unavailable timer only degrades to the CPU clock
+    # with a single warning.
+    global _warned_timer_unavailable
+    try:
+        backend = gpu_timer.get_backend()
+    except gpu_timer.GpuTimerUnavailable as exc:
+        if not _warned_timer_unavailable:
+            _warned_timer_unavailable = True
+            logger.warning(
+                "traceml_amd: GPU timing disabled for this run "
+                "(falling back to CPU wall clock): %s",
+                exc,
+            )
+        return None
     if backend is None:
         return None
     try:
@@ -159,6 +177,10 @@ def timed_region(name: str):
         close_event(event)
 
 
+# Guarded by _summary_lock: written at flush time on the training thread,
+# read by the rank-stats launch path (today also the training thread, but a
+# sampler-side reader must never see a half-updated dict).
+_summary_lock = threading.Lock()
 _last_cpu_summary: dict = {}
 
 #: event name -> rank-stats summary key (CPU wall clock, available at flush
@@ -174,7 +196,8 @@ _SUMMARY_KEYS = {
 
 
 def last_step_cpu_summary() -> dict:
-    return dict(_last_cpu_summary)
+    with _summary_lock:
+        return dict(_last_cpu_summary)
 
 
 def flush_step_time_buffer(step: int) -> None:
@@ -197,8 +220,9 @@ def flush_step_time_buffer(step: int) -> None:
                 summary[key] = summary.get(key, 0.0) + (
                     (event.cpu_end - event.cpu_start) * 1000.0
                 )
-        _last_cpu_summary.clear()
-        _last_cpu_summary.update(summary)
+        with _summary_lock:
+            _last_cpu_summary.clear()
+            _last_cpu_summary.update(summary)
     batch = StepTimeBatch(step=step, events=events, flushed_at=time.time())
     with _queue_lock:
         if len(_step_time_queue) >= STEP_TIME_QUEUE_MAX:

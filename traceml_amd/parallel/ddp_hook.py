@@ -61,7 +61,7 @@ def timed_allreduce_hook(state: _CommTimerState, bucket):
     if state.use_gpu:
         from traceml_amd.core import gpu_timer
 
-        backend = gpu_timer.get_backend()
+        backend = gpu_timer.get_backend_or_none()
         event = TimeEvent(
             name=event_names.DDP_COMM,
             device="cuda",

@@ -86,6 +86,17 @@ class TraceMLRuntime:
         self._stop.set()
         if self._thread is not None:
             self._thread.join(timeout=5.0)
+        # Settle the RCCL rank-stats exchange BEFORE the final sampler drain:
+        # waits bounded for in-flight gathers so none are pending when the
+        # process group is later destroyed (teardown-hang guard).
+        try:
+            from traceml_amd.parallel.rank_stats import (
+                disable_rank_stats_exchange,
+            )
+
+            disable_rank_stats_exchange(shutdown_timeout_sec=3.0)
+        except Exception:
+            logger.debug("traceml_amd: rank-stats shutdown failed", exc_info=True)
         # Final drain: give every sampler a last chance + publish + control.
         try:
             for sampler in self._samplers:

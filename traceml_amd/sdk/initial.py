@@ -210,6 +210,13 @@ def init(
         )
 
         try:
+            # Loud native-timer check happens HERE (before any training step),
+            # so a GPU box with a missing .so either refuses or runs no-op —
+            # it never kills a training step from inside the profiler
+            # (reference fail-open principle: architecture.md:54-59).
+            from traceml_amd.core import gpu_timer
+
+            gpu_timer.preflight_check()
             _start_runtime_for_init(
                 settings,
                 connect_timeout_sec,

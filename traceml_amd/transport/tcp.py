@@ -26,6 +26,10 @@ logger = logging.getLogger(__name__)
 
 LENGTH_PREFIX = struct.Struct(">I")
 MAX_FRAME_BYTES = 64 * 1024 * 1024
+#: Ingest queue bound — drop-oldest like every other queue in the pipeline
+#: (step-time handoff, SQLite ingest, stdout capture); if the aggregator
+#: loop stalls, fast clients must not grow host memory without limit.
+SERVER_QUEUE_MAX = 100_000
 
 
 class TCPServer:
@@ -40,6 +44,8 @@ class TCPServer:
         self._queue: deque = deque()
         self._data_event = threading.Event()
         self.port: Optional[int] = None
+        self.dropped = 0
+        self._last_drop_warn = 0.0
 
     def start(self) -> None:
         sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
@@ -118,6 +124,25 @@ class TCPServer:
             items = payload if isinstance(payload, list) else [payload]
             with self._queue_lock:
                 self._queue.extend(items)
+                overflow = len(self._queue) - SERVER_QUEUE_MAX
+                if overflow > 0:
+                    for _ in range(overflow):
+                        self._queue.popleft()
+                    self.dropped += overflow
+                    dropped = self.dropped
+                else:
+                    dropped = 0
+            if dropped:
+                import time as _time
+
+                now = _time.time()
+                if now - self._last_drop_warn > 10.0:
+                    self._last_drop_warn = now
+                    logger.warning(
+                        "traceml_amd: aggregator ingest queue full, "
+                        "dropped %d payload(s) total (oldest-first)",
+                        dropped,
+                    )
             self._data_event.set()
 
     def wait_for_data(self, timeout: float) -> bool:
