@@ -3,24 +3,38 @@
 
 The reference publishes no headline numbers — its design goal
 is "lightweight enough to leave on for a full production run". This bench
-MEASURES that: it trains the BASELINE.json workload (ResNet-50 bf16,
-synthetic data, random init) twice on N MI355X GPUs under DDP/RCCL —
-untraced, then fully traced (patches + trace_step + ring-stamp GPU timing +
-DDP-comm hook + rank-stats all-gather + runtime sampler + TCP aggregator +
-SQLite) — and reports the overhead percentage.
+MEASURES that: it trains a BASELINE.json workload (default: ResNet-50 bf16,
+synthetic data, random init) on N MI355X GPUs under DDP/RCCL — untraced and
+fully traced (patches + trace_step + ring-stamp GPU timing + DDP-comm hook +
+rank-stats all-gather + runtime sampler + TCP aggregator + SQLite),
+interleaved — and reports the overhead percentage.
+
+Workload arms (``--model``) cover every BASELINE.json config:
+  mlp       config 1 (toy MLP, CPU plumbing)
+  resnet50  config 2 (ResNet-50 bf16; default on GPU)
+  llama3    config 4 (Llama-3-8B bf16, HF TrainerCallback bracket path)
+  gpt2      config 5 (GPT-2 124M, Lightning callback hook path; optional
+            --creep-mb memory-creep injector)
+Config 3 (input straggler) is a diagnosis scenario, not an overhead arm —
+see examples/demo/ and profiles/.
 
 Contract (driver): ``python bench.py --gpus N --steps K --warmup W``; for
 N>1 launched under torch.distributed.run with one rank per GPU over RCCL.
-Each phase does W untimed warmup steps then times EXACTLY K steps bracketed
+Each arm does W untimed warmup steps then times EXACTLY K steps bracketed
 by barrier + torch.cuda.synchronize on both sides; the reported time is the
-MAX over ranks. Rank 0 prints ONE JSON line.
+MAX over ranks. Rank 0 prints ONE JSON line. The aggregator binds an
+EPHEMERAL port on rank 0 which is broadcast to the other ranks, so several
+benches can share one box without collisions.
 """
 
 from __future__ import annotations
 
 import argparse
+import contextlib
 import json
+import math
 import os
+import statistics
 import sys
 import time
 
@@ -30,16 +44,42 @@ if REPO_ROOT not in sys.path:
 
 import torch
 import torch.nn as nn
-from torch.utils.data import DataLoader, TensorDataset
+from torch.utils.data import DataLoader
+
+#: per-model defaults: (per-GPU batch, seq len) — image size for resnet
+MODEL_DEFAULTS = {
+    "resnet50": dict(batch=256, seq=None),
+    "mlp": dict(batch=32, seq=None),
+    "llama3": dict(batch=1, seq=4096),
+    "gpt2": dict(batch=8, seq=1024),
+}
+
+#: keep adding (traced, untraced) rounds until each phase has roughly this
+#: much cumulative timed wall time — pulls short driver runs (K=20) out of
+#: the DVFS noise floor without violating the K-steps-per-arm contract
+MIN_TIMED_SEC_PER_PHASE = 5.0
+MAX_ROUNDS = 9
 
 
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--steps", type=int, default=100)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--model", default="resnet50", choices=["resnet50", "mlp"])
-    p.add_argument("--batch", type=int, default=256, help="per-GPU batch size")
+    p.add_argument(
+        "--model",
+        default="resnet50",
+        choices=["resnet50", "mlp", "llama3", "gpt2"],
+    )
+    p.add_argument("--batch", type=int, default=None, help="per-GPU batch size")
+    p.add_argument("--seq", type=int, default=None, help="sequence length")
+    p.add_argument(
+        "--creep-mb",
+        type=float,
+        default=0.0,
+        help="gpt2 arm: retain this many MiB of fresh GPU tensors per step "
+        "(synthetic memory-creep injector, BASELINE config 5)",
+    )
     return p.parse_args()
 
 
@@ -75,42 +115,248 @@ class _PrebuiltBatches(torch.utils.data.Dataset):
         return self.batches[idx]
 
 
-def build_workload(args, device, use_gpu):
-    if args.model == "resnet50" and use_gpu:
+def _pin(t):
+    return t.pin_memory() if torch.cuda.is_available() else t
+
+
+class Workload:
+    """One BASELINE config arm: model + data + the step function.
+
+    ``bracket(model)`` returns the traced step context for this arm —
+    plain ``trace_step`` for the auto-patch arms, the HF TrainerCallback
+    bracket for llama3, the Lightning callback hooks for gpt2 — so each
+    arm measures its real integration path.
+    """
+
+    init_mode = "auto"
+    dtype = "bf16"
+
+    def __init__(self, args, device, use_gpu):
+        self.args = args
+        self.device = device
+        self.use_gpu = use_gpu
+        d = MODEL_DEFAULTS[args.model]
+        self.batch = args.batch or d["batch"]
+        self.seq = args.seq or d["seq"]
+        self.model = None
+        self.optimizer = None
+        self.loader = None
+
+    def bracket(self, model):
+        from traceml_amd.sdk.instrumentation import trace_step
+
+        return trace_step(model)
+
+    def _loader(self, batches):
+        return DataLoader(
+            _PrebuiltBatches(batches), batch_size=None, shuffle=False,
+            num_workers=0,
+        )
+
+    def config_json(self):
+        return {
+            "model": self.args.model,
+            "global_batch": self.batch,
+            "seq_len": self.seq,
+        }
+
+
+class ResNetWorkload(Workload):
+    """BASELINE config 2: ResNet-50 bf16, bs 256/GPU, synthetic ImageNet."""
+
+    def build(self):
         from traceml_amd.models.resnet import resnet50
 
         # channels_last: MIOpen picks NHWC implicit-GEMM kernels for bf16
         # instead of the naive NCHW fallbacks
-        model = resnet50().to(device).to(memory_format=torch.channels_last)
-        batch = args.batch
+        model = resnet50().to(self.device).to(memory_format=torch.channels_last)
         batches = [
             (
-                torch.randn(batch, 3, 224, 224)
-                .contiguous(memory_format=torch.channels_last)
-                .pin_memory(),
-                torch.randint(0, 1000, (batch,)).pin_memory(),
+                _pin(
+                    torch.randn(self.batch, 3, 224, 224).contiguous(
+                        memory_format=torch.channels_last
+                    )
+                ),
+                _pin(torch.randint(0, 1000, (self.batch,))),
             )
             for _ in range(4)
         ]
-        optimizer = torch.optim.SGD(
+        self.model = model
+        self.optimizer = torch.optim.SGD(
             model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4
         )
-    else:
-        # CPU fallback keeps the bench runnable without a GPU (plumbing check)
+        self.loader = self._loader(batches)
+        self.loss_fn = nn.CrossEntropyLoss()
+
+    def step(self, model, next_batch, autocast, trace_bracket):
+        x, y = next_batch()
+        ctx = trace_bracket(model) if trace_bracket else contextlib.nullcontext()
+        with ctx:
+            x = x.to(self.device, non_blocking=True)
+            y = y.to(self.device, non_blocking=True)
+            self.optimizer.zero_grad(set_to_none=True)
+            with autocast:
+                loss = self.loss_fn(model(x), y)
+            loss.backward()
+            self.optimizer.step()
+
+
+class MLPWorkload(ResNetWorkload):
+    """BASELINE config 1: toy MLP — keeps the bench runnable without a GPU."""
+
+    dtype = "fp32"
+
+    def build(self):
         from traceml_amd.models.mlp import TinyMLP
 
-        model = TinyMLP().to(device)
-        batch = 32
+        self.model = TinyMLP().to(self.device)
         batches = [
-            (torch.randn(batch, 256), torch.randint(0, 10, (batch,)))
+            (torch.randn(self.batch, 256), torch.randint(0, 10, (self.batch,)))
             for _ in range(4)
         ]
-        optimizer = torch.optim.SGD(model.parameters(), lr=0.01)
-    loader = DataLoader(
-        _PrebuiltBatches(batches), batch_size=None, shuffle=False, num_workers=0
-    )
-    loss_fn = nn.CrossEntropyLoss()
-    return model, optimizer, loader, loss_fn, batch
+        self.optimizer = torch.optim.SGD(self.model.parameters(), lr=0.01)
+        self.loader = self._loader(batches)
+        self.loss_fn = nn.CrossEntropyLoss()
+
+
+class LlamaWorkload(Workload):
+    """BASELINE config 4: Llama-3-8B bf16, HF integration path. The traced
+    bracket is the real ``TraceMLTrainerCallback`` (on_step_begin/on_step_end),
+    the same object an HF ``Trainer`` drives (integrations/huggingface.py).
+    On CPU the tiny same-architecture miniature keeps the arm runnable."""
+
+    def build(self):
+        from traceml_amd.models.llama import build_llama3
+
+        if self.use_gpu:
+            self.model = build_llama3(
+                tiny=False, seq_len=self.seq, device=self.device
+            )
+        else:
+            self.seq = min(self.seq or 128, 128)
+            self.model = build_llama3(tiny=True).to(self.device)
+        vocab = self.model.config.vocab_size
+        batches = [
+            (
+                _pin(torch.randint(0, vocab, (self.batch, self.seq))),
+                _pin(torch.randint(0, vocab, (self.batch, self.seq))),
+            )
+            for _ in range(4)
+        ]
+        self.loader = self._loader(batches)
+        self.optimizer = torch.optim.AdamW(
+            self.model.parameters(), lr=1e-5, weight_decay=0.0
+        )
+        from traceml_amd.integrations.huggingface import TraceMLTrainerCallback
+
+        self._callback = TraceMLTrainerCallback()
+
+    def bracket(self, model):
+        cb = self._callback
+
+        @contextlib.contextmanager
+        def hf_step():
+            cb.on_step_begin(None, None, None, model=model)
+            try:
+                yield
+            finally:
+                cb.on_step_end(None, None, None)
+
+        return hf_step()
+
+    def step(self, model, next_batch, autocast, trace_bracket):
+        ids, labels = next_batch()
+        ctx = trace_bracket(model) if trace_bracket else contextlib.nullcontext()
+        with ctx:
+            ids = ids.to(self.device, non_blocking=True)
+            labels = labels.to(self.device, non_blocking=True)
+            self.optimizer.zero_grad(set_to_none=True)
+            with autocast:
+                loss = model(input_ids=ids, labels=labels).loss
+            loss.backward()
+            self.optimizer.step()
+
+
+class GPT2Workload(Workload):
+    """BASELINE config 5: GPT-2 124M via the Lightning hook path. The traced
+    bracket drives the real ``TraceMLCallback`` in Lightning's documented
+    hook order (integrations/lightning.py): batch_start → wrapped forward →
+    before/after_backward → before_optimizer_step → before_zero_grad →
+    batch_end. Lightning itself is not in this image, so the hooks are
+    invoked directly — the integration code under test is identical.
+    ``--creep-mb`` retains fresh GPU tensors each step (creep injector)."""
+
+    init_mode = "manual"  # Lightning owns phase timing (manual mode)
+
+    def build(self):
+        from traceml_amd.models.gpt2 import GPT2
+
+        if self.use_gpu:
+            self.model = GPT2(max_seq=self.seq).to(self.device)
+        else:
+            self.seq = min(self.seq or 64, 64)
+            self.model = GPT2(
+                vocab_size=512, n_layers=2, d_model=64, n_heads=4,
+                max_seq=self.seq,
+            ).to(self.device)
+        vocab = self.model.tok.num_embeddings
+        batches = [
+            (
+                _pin(torch.randint(0, vocab, (self.batch, self.seq))),
+                _pin(torch.randint(0, vocab, (self.batch, self.seq))),
+            )
+            for _ in range(4)
+        ]
+        self.loader = self._loader(batches)
+        self.optimizer = torch.optim.AdamW(self.model.parameters(), lr=3e-4)
+        from traceml_amd.integrations.lightning import TraceMLCallback
+
+        self._callback = TraceMLCallback()
+        self._retained = []
+
+    def bracket(self, model):
+        cb = self._callback
+
+        @contextlib.contextmanager
+        def lightning_step():
+            cb.on_train_batch_start(None, model, None, 0)
+            try:
+                yield
+            finally:
+                cb.on_train_batch_end(None, model, None, None, 0)
+
+        return lightning_step()
+
+    def step(self, model, next_batch, autocast, trace_bracket):
+        ids, labels = next_batch()
+        cb = self._callback if trace_bracket else None
+        ctx = trace_bracket(model) if trace_bracket else contextlib.nullcontext()
+        with ctx:
+            ids = ids.to(self.device, non_blocking=True)
+            labels = labels.to(self.device, non_blocking=True)
+            with autocast:
+                loss = model(ids, labels=labels)["loss"]
+            if cb:
+                cb.on_before_backward(None, model, loss)
+            loss.backward()
+            if cb:
+                cb.on_after_backward(None, model)
+                cb.on_before_optimizer_step(None, model, self.optimizer)
+            self.optimizer.step()
+            if cb:
+                cb.on_before_zero_grad(None, model, self.optimizer)
+            self.optimizer.zero_grad(set_to_none=True)
+        if self.args.creep_mb > 0:
+            n = int(self.args.creep_mb * (1 << 20) // 4)
+            self._retained.append(torch.empty(n, device=self.device))
+
+
+WORKLOADS = {
+    "resnet50": ResNetWorkload,
+    "mlp": MLPWorkload,
+    "llama3": LlamaWorkload,
+    "gpt2": GPT2Workload,
+}
 
 
 def barrier_sync(world_size, use_gpu):
@@ -133,49 +379,58 @@ def max_over_ranks(value: float, world_size: int, use_gpu: bool) -> float:
     return float(t.item())
 
 
-def run_phase(
-    model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
-    traced: bool, trace_ctx=None,
-):
+def run_phase(work, model, args, world_size, use_gpu, traced: bool):
     """W warmup + K timed steps; returns wall seconds for the K steps (this
     rank). The step: fetch -> H2D -> forward -> loss -> backward -> step."""
-    import contextlib
-
     autocast = (
         torch.autocast("cuda", dtype=torch.bfloat16)
-        if use_gpu
+        if use_gpu and work.dtype == "bf16"
         else contextlib.nullcontext()
     )
-    it = iter(loader)
+    it = iter(work.loader)
 
     def next_batch():
         nonlocal it
         try:
             return next(it)
         except StopIteration:
-            it = iter(loader)
+            it = iter(work.loader)
             return next(it)
 
-    def one_step():
-        x, y = next_batch()
-        ctx = trace_ctx(model) if traced else contextlib.nullcontext()
-        with ctx:
-            x = x.to(device, non_blocking=True)
-            y = y.to(device, non_blocking=True)
-            optimizer.zero_grad(set_to_none=True)
-            with autocast:
-                loss = loss_fn(model(x), y)
-            loss.backward()
-            optimizer.step()
-
+    bracket = work.bracket if traced else None
     for _ in range(args.warmup):
-        one_step()
+        work.step(model, next_batch, autocast, bracket)
     barrier_sync(world_size, use_gpu)
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_step()
+        work.step(model, next_batch, autocast, bracket)
     barrier_sync(world_size, use_gpu)
     return time.perf_counter() - t0
+
+
+def negotiate_aggregator_port(rank, world_size, use_gpu, settings):
+    """Rank 0 starts the aggregator on an EPHEMERAL port (unless the env
+    pins one) and broadcasts the actual bound port, so concurrent benches
+    on a shared box never collide. Returns (aggregator_or_None, port)."""
+    aggregator = None
+    port = 0
+    if rank == 0:
+        from traceml_amd.aggregator.aggregator import TraceMLAggregator
+
+        if "TRACEML_AGGREGATOR_PORT" not in os.environ:
+            settings.aggregator_port = 0  # ephemeral
+        aggregator = TraceMLAggregator(settings)
+        aggregator.start()
+        port = int(aggregator.port)
+    if world_size > 1:
+        import torch.distributed as dist
+
+        box = [port]
+        dist.broadcast_object_list(box, src=0)
+        port = int(box[0])
+    settings.aggregator_port = port
+    os.environ["TRACEML_AGGREGATOR_PORT"] = str(port)
+    return aggregator, port
 
 
 def main():
@@ -187,53 +442,44 @@ def main():
         # MIOpen find-mode kernel selection: tune once in warmup, then stable
         torch.backends.cudnn.benchmark = True
 
-    model, optimizer, loader, loss_fn, batch = build_workload(args, device, use_gpu)
+    if args.model == "resnet50" and not use_gpu:
+        args.model = "mlp"  # CPU fallback keeps the default runnable GPU-free
+    work = WORKLOADS[args.model](args, device, use_gpu)
+    work.build()
+    model = work.model
     if world_size > 1:
         from torch.nn.parallel import DistributedDataParallel as DDP
 
         model = DDP(model, device_ids=[local_rank] if use_gpu else None)
 
-    # ---- phase 1: untraced baseline (first pass absorbs cold-GPU effects:
-    # MIOpen find, allocator growth, DVFS ramp; it is re-measured
-    # interleaved with the traced phase below and the MIN is used) ----
-    t_off = run_phase(
-        model, optimizer, loader, loss_fn, device, args, world_size, use_gpu,
-        traced=False,
-    )
+    # ---- burn-in arm (absorbs cold-GPU effects: MIOpen find, allocator
+    # growth, DVFS ramp); re-measured interleaved below, MIN never used ----
+    t_burn = run_phase(work, model, args, world_size, use_gpu, traced=False)
 
     # ---- bring up the full tracing stack ----
     logs_dir = os.path.join(REPO_ROOT, "gpurun_out", "bench_logs")
-    session_id = f"bench-{os.environ.get('MASTER_PORT', '0')}-{world_size}"
+    session_id = f"bench-{args.model}-{os.getpid() if world_size == 1 else os.environ.get('MASTER_PORT', '0')}-{world_size}"
     os.environ["TRACEML_LOGS_DIR"] = logs_dir
     os.environ["TRACEML_SESSION_ID"] = session_id
     os.environ["TRACEML_INTERVAL"] = "1.0"
     os.environ["TRACEML_FINALIZE_TIMEOUT"] = "30"  # bench must exit promptly
-    os.environ["TRACEML_AGGREGATOR_PORT"] = os.environ.get(
-        "TRACEML_AGGREGATOR_PORT", "29877"
-    )
     os.environ["TRACEML_EXPECTED_RANKS"] = str(world_size)
 
     from traceml_amd.runtime.settings import TraceMLSettings
 
     settings = TraceMLSettings.from_env()
-    aggregator = None
-    if rank == 0:
-        from traceml_amd.aggregator.aggregator import TraceMLAggregator
-
-        aggregator = TraceMLAggregator(settings)
-        aggregator.start()
-    if world_size > 1:
-        import torch.distributed as dist
-
-        dist.barrier()
+    aggregator, _port = negotiate_aggregator_port(
+        rank, world_size, use_gpu, settings
+    )
 
     import traceml_amd
     from traceml_amd.runtime import lifecycle
-    from traceml_amd.sdk.instrumentation import trace_step
 
     handle = lifecycle.start_runtime(settings, fail_open=False,
                                      register_atexit=False)
-    traceml_amd.init(aggregator_port=settings.aggregator_port)
+    traceml_amd.init(
+        mode=work.init_mode, aggregator_port=settings.aggregator_port
+    )
     if world_size > 1:
         from traceml_amd.parallel.ddp_hook import attach_ddp_comm_timing
         from traceml_amd.parallel.rank_stats import enable_rank_stats_exchange
@@ -243,24 +489,21 @@ def main():
 
     # ---- interleaved measurement: alternate (on, off) rounds and compare
     # MEDIANS, so DVFS/thermal/dataloader jitter (±2-3% on this workload)
-    # does not masquerade as instrumentation overhead. The first `t_off`
-    # above is a discarded extra burn-in arm. ----
-    import statistics
+    # does not masquerade as instrumentation overhead. Rounds scale up
+    # until each phase has ≥MIN_TIMED_SEC_PER_PHASE of cumulative timed
+    # wall time (short driver runs would otherwise sit inside the DVFS
+    # noise floor). The round count is derived from the burn-in arm, which
+    # is identical on every rank up to jitter; it is then MAX-reduced so
+    # all ranks run the same number of rounds. ----
+    rounds = max(
+        3, min(MAX_ROUNDS, math.ceil(MIN_TIMED_SEC_PER_PHASE / max(t_burn, 1e-3)))
+    )
+    rounds = int(max_over_ranks(float(rounds), world_size, use_gpu))
 
     offs, ons = [], []
-    for _ in range(3):
-        ons.append(
-            run_phase(
-                model, optimizer, loader, loss_fn, device, args, world_size,
-                use_gpu, traced=True, trace_ctx=trace_step,
-            )
-        )
-        offs.append(
-            run_phase(
-                model, optimizer, loader, loss_fn, device, args, world_size,
-                use_gpu, traced=False,
-            )
-        )
+    for _ in range(rounds):
+        ons.append(run_phase(work, model, args, world_size, use_gpu, traced=True))
+        offs.append(run_phase(work, model, args, world_size, use_gpu, traced=False))
     t_off_med = statistics.median(offs)
     t_on_med = statistics.median(ons)
     #: same-arm spread = measurement noise floor for this config
@@ -270,6 +513,10 @@ def main():
     t_off = max_over_ranks(t_off_med, world_size, use_gpu)
     t_on = max_over_ranks(t_on_med, world_size, use_gpu)
     noise_pct = max_over_ranks(noise_pct, world_size, use_gpu)
+
+    from traceml_amd.sdk.instrumentation import self_overhead_us_per_step
+
+    self_us = self_overhead_us_per_step()
 
     # ---- teardown + diagnosis (outside the timed regions) ----
     handle.stop()
@@ -301,21 +548,28 @@ def main():
             "higher_is_better": False,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": work.dtype if use_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": batch * world_size,
-                "seq_len": None,
+                "global_batch": work.batch * world_size,
+                "seq_len": work.seq,
                 "image_size": 224 if args.model == "resnet50" else None,
                 "parallelism": f"dp{world_size}",
                 "ms_per_step_untraced": ms_plain,
                 "noise_pct_same_arm_spread": noise_pct,
-                "rounds_per_arm": 3,
+                "rounds_per_arm": rounds,
+                "self_overhead_us_per_step": self_us,
                 "diagnosis": diagnosis,
                 "tracing": "full stack: patches + trace_step + hip ring stamps"
                 " + ddp_comm hook + rccl rank stats + sampler thread + TCP"
                 " aggregator + SQLite",
+                "integration_path": {
+                    "resnet50": "auto patches",
+                    "mlp": "auto patches",
+                    "llama3": "HF TraceMLTrainerCallback bracket",
+                    "gpt2": "Lightning TraceMLCallback hooks (manual mode)",
+                }[args.model],
             },
         }
         print(json.dumps(result), flush=True)

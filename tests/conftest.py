@@ -17,6 +17,17 @@ def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
 
 
+def free_port() -> int:
+    """An OS-assigned free TCP port. Fixed port numbers collide when several
+    test runs (or the driver's scale bench) share one box; every subprocess
+    test that needs a rendezvous or aggregator port should use this."""
+    import socket
+
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 @pytest.fixture(autouse=True)
 def _clean_traceml_state():
     """Every test starts with pristine patches/queues/config and no TRACEML_*

@@ -11,6 +11,8 @@ import textwrap
 
 import pytest
 
+from tests.conftest import free_port
+
 REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 WORKER = textwrap.dedent(
@@ -95,7 +97,7 @@ def test_ddp_hook_and_rank_stats_gloo_ws2(tmp_path):
             "--nnodes=1",
             "--nproc-per-node=2",
             "--master-addr=127.0.0.1",
-            "--master-port=29611",
+            "--master-port=%d" % free_port(),
             str(worker),
         ],
         env=env,
@@ -148,7 +150,7 @@ def test_ddp_gradients_still_correct_with_timing_hook(tmp_path):
             "--nnodes=1",
             "--nproc-per-node=2",
             "--master-addr=127.0.0.1",
-            "--master-port=29612",
+            "--master-port=%d" % free_port(),
             str(script),
         ],
         env=env,

@@ -8,6 +8,8 @@ import time
 
 import pytest
 
+from tests.conftest import free_port
+
 torch = pytest.importorskip("torch")
 
 pytestmark = pytest.mark.gpu
@@ -245,7 +247,7 @@ def test_fsdp_wrap_traced_on_gpu(armed_auto_config):
         import os
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29630")
+        os.environ["MASTER_PORT"] = str(free_port())
         dist.init_process_group("nccl", rank=0, world_size=1)
     from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
 
@@ -363,7 +365,7 @@ def test_ddp_comm_hook_on_rccl_ws1(armed_auto_config):
 
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29631")
+        os.environ["MASTER_PORT"] = str(free_port())
         dist.init_process_group("nccl", rank=0, world_size=1)
     from torch.nn.parallel import DistributedDataParallel as DDP
 
@@ -416,7 +418,7 @@ def test_rank_stats_gather_on_rccl_ws1():
 
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29633")
+        os.environ["MASTER_PORT"] = str(free_port())
         dist.init_process_group("nccl", rank=0, world_size=1)
     from traceml_amd.core import timing
     from traceml_amd.parallel.rank_stats import FIELDS, RankStatsExchange

@@ -12,6 +12,8 @@ import time
 
 import pytest
 
+from tests.conftest import free_port
+
 REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 USER_SCRIPT = textwrap.dedent(
@@ -43,7 +45,7 @@ USER_SCRIPT = textwrap.dedent(
 
 @pytest.mark.timeout(300)
 def test_serve_and_direct_init(tmp_path):
-    port = "29891"
+    port = str(free_port())
     env = dict(os.environ)
     env.update(
         {
@@ -140,7 +142,7 @@ MIDRUN_SCRIPT = textwrap.dedent(
 def test_midrun_summary_rpc(tmp_path):
     """traceml_amd.summary() during the run: file-RPC to the live
     aggregator produces a flat tracker dict before the run ends."""
-    port = "29892"
+    port = str(free_port())
     env = dict(os.environ)
     env.update(
         {

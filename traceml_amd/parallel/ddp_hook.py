@@ -95,10 +95,18 @@ def timed_allreduce_hook(state: _CommTimerState, bucket):
 
 
 def attach_ddp_comm_timing(ddp_model, process_group=None) -> Optional[_CommTimerState]:
-    """Register the timing comm hook on a DistributedDataParallel model."""
+    """Register the timing comm hook on a DistributedDataParallel model.
+
+    Idempotent: DDP allows exactly one comm hook per model, and both an
+    explicit call (bench/user) and trace_step's auto-attach may race to be
+    first — a second attach returns the existing state silently."""
+    existing = getattr(ddp_model, "_traceml_comm_timer_state", None)
+    if existing is not None:
+        return existing
     try:
         state = _CommTimerState(process_group)
         ddp_model.register_comm_hook(state, timed_allreduce_hook)
+        ddp_model._traceml_comm_timer_state = state
         return state
     except Exception:
         logger.warning("traceml_amd: could not attach DDP comm timing", exc_info=True)
