@@ -51,69 +51,136 @@ class CLIDisplayDriver(DisplayDriver):
         from rich.table import Table
         from rich.console import Group
 
+        from traceml_amd.renderers.comm import load_latest_gather, render_comm
+        from traceml_amd.renderers.step_memory import render_step_memory
+        from traceml_amd.renderers.step_time import render_step_time
+        from traceml_amd.renderers.system import render_system
         from traceml_amd.steptime.pipeline import LiveStepTimeSession
 
         if self._session is None:
             self._session = LiveStepTimeSession(db_path)
         result, freshness = self._session.tick()
-        window = result.window
-        diag = result.diagnosis.primary
+        st = render_step_time(result.window, result.diagnosis)
+        diag = st["diagnosis"]
 
         header = (
-            f"[{_SEVERITY_STYLE.get(diag.severity, 'cyan')}]"
-            f"{diag.status}[/]  {diag.summary}"
+            f"[{_SEVERITY_STYLE.get(diag.get('severity'), 'cyan')}]"
+            f"{diag.get('status')}[/]  {diag.get('summary')}"
         )
         if freshness not in ("live", "cold"):
             header += f"  [dim]({freshness})[/dim]"
         renderables = [header]
 
-        if window.has_data:
+        if st["available"]:
             table = Table(title=None, expand=False, pad_edge=False)
             table.add_column("metric")
-            for rank in window.ranks_used:
-                table.add_column(f"r{rank}", justify="right")
-            for metric, label in (
-                ("step_time_ms", "step (ms)"),
-                ("input_wait_ms", "input"),
-                ("h2d_ms", "h2d"),
-                ("forward_ms", "forward"),
-                ("backward_ms", "backward"),
-                ("optimizer_ms", "optimizer"),
-                ("ddp_comm_ms", "ddp comm"),
-                ("residual_ms", "residual"),
-            ):
-                row = [label]
-                any_value = False
-                for rank in window.ranks_used:
-                    value = window.ranks[rank].get(metric)
-                    row.append("—" if value is None else f"{value:.1f}")
-                    any_value = any_value or value is not None
-                if any_value:
-                    table.add_row(*row)
-            renderables.append(table)
-            renderables.append(
-                f"[dim]{window.steps_analyzed} aligned steps · "
-                f"{window.clock} clock · strategy {window.training_strategy}[/dim]"
-            )
-        # cross-section findings (memory/system/process), top 3 actionable
-        try:
-            from traceml_amd.renderers.views import issues_view
-
-            neutral = {"NORMAL", "BALANCED", "NO_DATA", "NO_GPU", "WARMUP"}
-            findings = [
-                i
-                for i in issues_view(db_path, result.diagnosis)
-                if i["kind"] not in neutral and i["section"] != "step_time"
-            ][:3]
-            for issue in findings:
-                style = _SEVERITY_STYLE.get(issue["severity"], "cyan")
-                renderables.append(
-                    f"[{style}]{issue['status']}[/] "
-                    f"[dim]\\[{issue['section']}][/dim] {issue['summary']}"
+            cohorts = st["cohorts"]
+            for rank in st["ranks"]:
+                mark = {"slow": " [red]▲[/red]", "fast": " [cyan]▼[/cyan]"}.get(
+                    cohorts.get(rank), ""
                 )
+                table.add_column(f"r{rank}{mark}", justify="right")
+            for row in st["rows"]:
+                table.add_row(
+                    row["label"],
+                    *[
+                        row["cells"][rank]["text"] or "—"
+                        for rank in st["ranks"]
+                    ],
+                )
+            renderables.append(table)
+            skew = st.get("skew")
+            if skew and skew["skew_fraction"] > 0.05:
+                renderables.append(
+                    f"[yellow]rank skew[/yellow] r{skew['worst_rank']} "
+                    f"{skew['worst_ms']:.1f} ms vs median "
+                    f"{skew['median_ms']:.1f} ms "
+                    f"(+{skew['skew_fraction'] * 100:.0f}%)"
+                )
+            footer = st["footer"]
+            renderables.append(
+                f"[dim]{footer['steps_analyzed']} aligned steps · "
+                f"{footer['clock']} clock · strategy {footer['strategy']}[/dim]"
+            )
+        try:
+            self._append_side_sections(
+                renderables, db_path, result,
+                render_step_memory, render_system, render_comm,
+                load_latest_gather,
+            )
         except Exception:
-            pass
+            logger.debug("traceml_amd: cli side sections failed", exc_info=True)
         return Panel(Group(*renderables), title="traceml-amd live", border_style="blue")
+
+    def _append_side_sections(
+        self, renderables, db_path, result,
+        render_step_memory, render_system, render_comm, load_latest_gather,
+    ) -> None:
+        from traceml_amd.diagnostics.step_memory.api import load_memory_series
+        from traceml_amd.diagnostics.system.api import load_system_context
+
+        band_style = {"ok": "green", "warn": "yellow", "crit": "bold red",
+                      "low": "yellow", "moderate": "cyan"}
+
+        mem = render_step_memory(load_memory_series(db_path))
+        if mem["available"]:
+            parts = []
+            for card in mem["cards"]:
+                if card["peak_alloc_bytes"] is None:
+                    continue
+                style = band_style.get(card["pressure_band"] or "ok", "green")
+                pct = (
+                    f" ({card['pressure_fraction'] * 100:.0f}%)"
+                    if card["pressure_fraction"] is not None
+                    else ""
+                )
+                parts.append(
+                    f"r{card['rank']} [{style}]"
+                    f"{card['peak_alloc_gib']}G/"
+                    f"{card['peak_reserved_gib']}G[/]{pct}"
+                )
+            if parts:
+                renderables.append("[dim]mem alloc/reserved:[/dim] " + "  ".join(parts))
+
+        sys_view = render_system(load_system_context(db_path))
+        if sys_view["available"] and sys_view["gpus"]:
+            parts = []
+            for g in sys_view["gpus"]:
+                util = "—" if g["util_percent"] is None else f"{g['util_percent']:.0f}%"
+                style = band_style.get(g["util_band"] or "ok", "green")
+                temp = "" if g["temp_c"] is None else f" {g['temp_c']:.0f}°C"
+                parts.append(f"gpu{g['gpu']} [{style}]{util}[/]{temp}")
+            renderables.append("[dim]gpus:[/dim] " + "  ".join(parts))
+
+        comm = render_comm(load_latest_gather(db_path))
+        if comm["available"]:
+            line = (
+                f"[dim]xGMI rank stats:[/dim] gather "
+                f"{comm['gather_latency_ms']:.2f} ms"
+                if comm["gather_latency_ms"] is not None
+                else "[dim]xGMI rank stats[/dim]"
+            )
+            if comm.get("step_skew"):
+                line += f" · step spread {comm['step_skew']['spread_ms']:.1f} ms"
+            if comm.get("slowest_rank") is not None:
+                line += f" · slowest r{comm['slowest_rank']}"
+            renderables.append(line)
+
+        # cross-section findings (memory/system/process), top 3 actionable
+        from traceml_amd.renderers.views import issues_view
+
+        neutral = {"NORMAL", "BALANCED", "NO_DATA", "NO_GPU", "WARMUP"}
+        findings = [
+            i
+            for i in issues_view(db_path, result.diagnosis)
+            if i["kind"] not in neutral and i["section"] != "step_time"
+        ][:3]
+        for issue in findings:
+            style = _SEVERITY_STYLE.get(issue["severity"], "cyan")
+            renderables.append(
+                f"[{style}]{issue['status']}[/] "
+                f"[dim]\\[{issue['section']}][/dim] {issue['summary']}"
+            )
 
     def stop(self) -> None:
         if self._live is not None:

@@ -139,6 +139,8 @@ class RankStatsWriter(ProjectionWriter):
             [
                 ("timestamp", "REAL"),
                 ("world_size_gathered", "INTEGER"),
+                ("gather_latency_ms", "REAL"),
+                ("gather_latency_ms_mean", "REAL"),
                 ("ranks_json", "TEXT"),
             ],
         )

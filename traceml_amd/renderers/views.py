@@ -143,7 +143,10 @@ def history_view(window, max_points: int = 120) -> dict:
 
 def live_view(db_path: str, session=None) -> dict:
     """One payload for live surfaces; optionally freshness-bridged via a
-    LiveStepTimeSession."""
+    LiveStepTimeSession. ``sections`` carries the per-section renderer view
+    models (renderers/{step_time,step_memory,system,process,comm}.py) that
+    the CLI, dashboard and HTML report all share; the flat top-level keys
+    are kept for backward compatibility."""
     if session is not None:
         result, freshness = session.tick()
     else:
@@ -151,8 +154,40 @@ def live_view(db_path: str, session=None) -> dict:
 
         result = StepTimePipeline(db_path, profile="live").run()
         freshness = "live" if result.window.has_data else "cold"
+
+    from traceml_amd.diagnostics.process.api import (
+        diagnose_process,
+        load_process_context,
+    )
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+    from traceml_amd.diagnostics.system.api import (
+        diagnose_system,
+        load_system_context,
+    )
+    from traceml_amd.renderers.comm import load_latest_gather, render_comm
+    from traceml_amd.renderers.process import render_process
+    from traceml_amd.renderers.step_memory import render_step_memory
+    from traceml_amd.renderers.step_time import render_step_time
+    from traceml_amd.renderers.system import render_system
+
+    memory_series = load_memory_series(db_path)
+    system_ctx = load_system_context(db_path)
+    process_ctx = load_process_context(db_path)
+    sections = {
+        "step_time": render_step_time(result.window, result.diagnosis),
+        "step_memory": render_step_memory(
+            memory_series, diagnose_step_memory(memory_series)
+        ),
+        "system": render_system(system_ctx, diagnose_system(system_ctx)),
+        "process": render_process(process_ctx, diagnose_process(process_ctx)),
+        "comm": render_comm(load_latest_gather(db_path)),
+    }
     payload = {
         "freshness": freshness,
+        "sections": sections,
         "step_time": step_time_view(result.window, result.diagnosis),
         "memory": memory_view(db_path),
         "system": system_view(db_path),
