@@ -1,220 +1,176 @@
-"""``traceml-amd compare A.json B.json`` (reference: reporting/compare/*, ~2k LoC).
+"""``traceml-amd compare A.json B.json`` (reference: reporting/compare/*,
+~2k LoC: io.py:200 schema-versioned readers, sections/, policy.py:216,
+verdict.py:506 rule chain).
 
-Schema-version-aware reader, per-metric significance classification, and a
-rule-chain verdict: REGRESSION / IMPROVEMENT / NEUTRAL / MIXED.
+Pipeline: strict schema-versioned load → normalize → per-section comparers
+(step_time clock-aware, step_memory, system, process) → priority rule-chain
+verdict → sectioned text render. ``--fail-on-regression`` turns the verdict
+into a CI gate (exit 4).
 """
 
 from __future__ import annotations
 
-import json
 import sys
-from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
-from traceml_amd.reporting.compare import policy
+from traceml_amd.reporting.compare import io as compare_io
+from traceml_amd.reporting.compare.model import CompareSection
+from traceml_amd.reporting.compare.sections import (
+    compare_process,
+    compare_step_memory,
+    compare_step_time,
+    compare_system,
+    per_rank_step_time,
+)
+from traceml_amd.reporting.compare.verdict import decide_verdict
 
-_COMPARED_METRICS: List[Tuple[str, str, str]] = [
-    # (section, metric, unit)
-    ("step_time", "step_time_ms", "ms"),
-    ("step_time", "input_wait_ms", "ms"),
-    ("step_time", "h2d_ms", "ms"),
-    ("step_time", "compute_ms", "ms"),
-    ("step_time", "forward_ms", "ms"),
-    ("step_time", "backward_ms", "ms"),
-    ("step_time", "optimizer_ms", "ms"),
-    ("step_time", "ddp_comm_ms", "ms"),
-    ("step_time", "residual_ms", "ms"),
-    ("step_memory", "peak_allocated_bytes", "bytes"),
-    ("step_memory", "peak_reserved_bytes", "bytes"),
-]
-
-
-@dataclass
-class MetricDelta:
-    section: str
-    metric: str
-    unit: str
-    baseline: Optional[float]
-    candidate: Optional[float]
-    status: str  # REGRESSION | IMPROVEMENT | NEUTRAL | INCOMPARABLE
-
-    @property
-    def delta(self) -> Optional[float]:
-        if self.baseline is None or self.candidate is None:
-            return None
-        return self.candidate - self.baseline
-
-    @property
-    def pct(self) -> Optional[float]:
-        if self.delta is None or not self.baseline:
-            return None
-        return self.delta / self.baseline * 100.0
-
-
-def load_summary(path: str) -> dict:
-    with open(path, "r", encoding="utf-8") as f:
-        payload = json.load(f)
-    version = payload.get("schema_version")
-    if version is None:
-        raise ValueError(f"{path}: not a traceml final summary (no schema_version)")
-    return payload
-
-
-def _metric_value(payload: dict, section: str, metric: str) -> Optional[float]:
-    value = payload.get(section, {}).get("global", {}).get("average", {}).get(metric)
-    return float(value) if isinstance(value, (int, float)) else None
-
-
-def _classify(delta: Optional[float], baseline: Optional[float], unit: str) -> str:
-    if delta is None:
-        return "INCOMPARABLE"
-    floor = (
-        policy.ABSOLUTE_BYTES_FLOOR if unit == "bytes" else policy.ABSOLUTE_MS_FLOOR
-    )
-    if abs(delta) < floor:
-        return "NEUTRAL"
-    if baseline and abs(delta / baseline) < policy.RELATIVE_SIGNIFICANCE:
-        return "NEUTRAL"
-    return "REGRESSION" if delta > 0 else "IMPROVEMENT"
+#: kept for callers of the old reader API
+load_summary = compare_io.load_summary
 
 
 def compare_payloads(baseline: dict, candidate: dict) -> dict:
-    deltas: List[MetricDelta] = []
-    for section, metric, unit in _COMPARED_METRICS:
-        b = _metric_value(baseline, section, metric)
-        c = _metric_value(candidate, section, metric)
-        delta = (c - b) if (b is not None and c is not None) else None
-        deltas.append(
-            MetricDelta(section, metric, unit, b, c, _classify(delta, b, unit))
-        )
+    """Full sectioned comparison of two loaded final summaries."""
+    baseline, notes_a = compare_io.normalize_summary(dict(baseline))
+    candidate, notes_b = compare_io.normalize_summary(dict(candidate))
+    sections: Dict[str, CompareSection] = {
+        "step_time": compare_step_time(baseline, candidate),
+        "step_memory": compare_step_memory(baseline, candidate),
+        "system": compare_system(baseline, candidate),
+        "process": compare_process(baseline, candidate),
+    }
+    finding = decide_verdict(sections)
 
-    statuses = {d.status for d in deltas if d.status != "INCOMPARABLE"}
-    headline_metric = next(
-        (d for d in deltas if d.metric == "step_time_ms"), None
-    )
-    if "REGRESSION" in statuses and "IMPROVEMENT" in statuses:
-        verdict = "MIXED"
-    elif "REGRESSION" in statuses:
-        verdict = "REGRESSION"
-    elif "IMPROVEMENT" in statuses:
-        verdict = "IMPROVEMENT"
-    elif statuses:
-        verdict = "NEUTRAL"
-    else:
-        verdict = "INCOMPARABLE"
-    # headline step-time movement dominates the verdict when significant
-    if headline_metric and headline_metric.status in ("REGRESSION", "IMPROVEMENT"):
-        verdict = headline_metric.status
+    # flat list kept for existing consumers (CI scripts, old tests)
+    flat: List[dict] = []
+    for name in ("step_time", "step_memory"):
+        for key, metric in sections[name].metrics.items():
+            p = metric.to_payload()
+            flat.append(
+                {
+                    "section": name,
+                    "metric": key,
+                    "unit": p["unit"],
+                    "baseline": p["baseline"],
+                    "candidate": p["candidate"],
+                    "delta": p["delta"],
+                    "pct": p["pct"],
+                    "status": p["status"],
+                    "significance": p["significance"],
+                }
+            )
 
+    st_diag = sections["step_time"].diagnosis
     return {
-        "verdict": verdict,
+        "verdict": finding.verdict,
+        "finding": finding.to_payload(),
         "baseline_diagnosis": baseline.get("primary_diagnosis", {}).get("kind"),
-        "candidate_diagnosis": candidate.get("primary_diagnosis", {}).get("kind"),
-        "diagnosis_transition": _diagnosis_transition(baseline, candidate),
-        "per_rank_step_time": _per_rank_compare(
-            baseline, candidate, "step_time", "step_time_ms"
+        "candidate_diagnosis": candidate.get("primary_diagnosis", {}).get(
+            "kind"
         ),
-        "metrics": [
-            {
-                "section": d.section,
-                "metric": d.metric,
-                "unit": d.unit,
-                "baseline": d.baseline,
-                "candidate": d.candidate,
-                "delta": d.delta,
-                "pct": d.pct,
-                "status": d.status,
-            }
-            for d in deltas
-        ],
+        "diagnosis_transition": (
+            st_diag.to_payload() if st_diag else None
+        ),
+        "sections": {k: s.to_payload() for k, s in sections.items()},
+        "per_rank_step_time": per_rank_step_time(baseline, candidate),
+        "metrics": flat,
+        "notes": notes_a + notes_b,
     }
 
 
-_SEVERITY_RANK = {"info": 0, "warn": 1, "crit": 2}
+def _fmt_value(value: Optional[float], unit: str) -> str:
+    if value is None:
+        return "—"
+    if unit == "bytes":
+        return f"{value / (1 << 30):.2f}G"
+    if unit == "percent":
+        return f"{value:.1f}"
+    return f"{value:.1f}"
 
 
-def _diagnosis_transition(baseline: dict, candidate: dict) -> dict:
-    b = baseline.get("primary_diagnosis", {})
-    c = candidate.get("primary_diagnosis", {})
-    b_sev = _SEVERITY_RANK.get(b.get("severity"), 0)
-    c_sev = _SEVERITY_RANK.get(c.get("severity"), 0)
-    if c_sev < b_sev:
-        direction = "resolved" if c_sev == 0 else "improved"
-    elif c_sev > b_sev:
-        direction = "worsened"
-    elif b.get("kind") != c.get("kind"):
-        direction = "changed"
-    else:
-        direction = "unchanged"
-    return {
-        "from": {"kind": b.get("kind"), "severity": b.get("severity")},
-        "to": {"kind": c.get("kind"), "severity": c.get("severity")},
-        "direction": direction,
-    }
-
-
-def _per_rank_compare(
-    baseline: dict, candidate: dict, section: str, metric: str
-) -> List[dict]:
-    def rank_values(payload):
-        rows = payload.get(section, {}).get("groups", {}).get("rows", {})
-        return {
-            key: row.get("metrics", {}).get(metric)
-            for key, row in rows.items()
-        }
-
-    b_rows = rank_values(baseline)
-    c_rows = rank_values(candidate)
-    out = []
-    for key in sorted(set(b_rows) | set(c_rows), key=lambda k: (len(k), k)):
-        b = b_rows.get(key)
-        c = c_rows.get(key)
-        delta = (c - b) if (b is not None and c is not None) else None
-        out.append(
-            {
-                "rank": key,
-                "baseline": b,
-                "candidate": c,
-                "delta": delta,
-                "pct": (delta / b * 100.0) if (delta is not None and b) else None,
-            }
-        )
+def _fmt_delta(m: dict) -> str:
+    if m["delta"] is None:
+        return "—"
+    sign = "+" if m["delta"] >= 0 else ""
+    out = f"{sign}{_fmt_value(m['delta'], m['unit'])}"
+    if m.get("pct") is not None:
+        out += f" ({sign}{m['pct']:.1f}%)"
     return out
 
 
-def render_compare(result: dict) -> str:
-    transition = result.get("diagnosis_transition", {})
+_SECTION_TITLES = {
+    "step_time": "Step time",
+    "step_memory": "Step memory",
+    "system": "System (context)",
+    "process": "Process (context)",
+}
+
+
+def render_compare(result: dict, label_a: str = "A", label_b: str = "B") -> str:
+    finding = result.get("finding", {})
+    transition = result.get("diagnosis_transition") or {}
     lines = [
-        f"TraceML-AMD Compare Verdict: {result['verdict']}",
+        f"TraceML-AMD Compare Verdict: {result['verdict']}"
+        + (
+            f"  [{finding['significance']}]"
+            if finding.get("significance")
+            else ""
+        ),
+        f"  {finding.get('title', '')}: {finding.get('detail', '')}",
+        f"  runs: {label_a} (baseline) vs {label_b} (candidate)",
         f"  diagnosis: {result['baseline_diagnosis']} -> "
         f"{result['candidate_diagnosis']}"
-        + (f"  ({transition.get('direction')})" if transition else ""),
-        "",
-        f"  {'metric':<28} {'baseline':>12} {'candidate':>12} {'delta':>14}  status",
+        + (
+            f"  ({transition.get('direction')})"
+            if transition.get("direction")
+            else ""
+        ),
     ]
-    for m in result["metrics"]:
-        if m["baseline"] is None and m["candidate"] is None:
+    for note in result.get("notes", []):
+        lines.append(f"  note: {note}")
+
+    for name, title in _SECTION_TITLES.items():
+        section = result.get("sections", {}).get(name)
+        if not section:
             continue
+        metrics = [
+            dict(m, metric=key)
+            for key, m in section.get("metrics", {}).items()
+            if m["baseline"] is not None or m["candidate"] is not None
+        ]
+        if not metrics and not section.get("notes"):
+            continue
+        lines.append("")
+        diag = section.get("diagnosis_transition") or {}
+        diag_txt = ""
+        if diag.get("from", {}).get("kind") or diag.get("to", {}).get("kind"):
+            diag_txt = (
+                f"   [{diag['from'].get('kind')} -> {diag['to'].get('kind')}"
+                f", {diag.get('direction')}]"
+            )
+        lines.append(f"  {title}{diag_txt}")
+        for note in section.get("notes", []):
+            lines.append(f"    note: {note}")
+        if metrics:
+            lines.append(
+                f"    {'metric':<30} {'baseline':>12} {'candidate':>12} "
+                f"{'delta':>18}  status"
+            )
+        for m in metrics:
+            status = m["status"]
+            if m.get("significance") in ("moderate", "material") and status in (
+                "REGRESSION", "IMPROVEMENT",
+            ):
+                status += f" ({m['significance']})"
+            lines.append(
+                f"    {m.get('label') or m['metric']:<30} "
+                f"{_fmt_value(m['baseline'], m['unit']):>12} "
+                f"{_fmt_value(m['candidate'], m['unit']):>12} "
+                f"{_fmt_delta(m):>18}  {status}"
+            )
 
-        def fmt(v):
-            if v is None:
-                return "—"
-            if m["unit"] == "bytes":
-                return f"{v / (1 << 30):.2f}G"
-            return f"{v:.1f}"
-
-        delta_s = "—"
-        if m["delta"] is not None:
-            sign = "+" if m["delta"] >= 0 else ""
-            delta_s = f"{sign}{fmt(m['delta'])}"
-            if m["pct"] is not None:
-                delta_s += f" ({sign}{m['pct']:.1f}%)"
-        lines.append(
-            f"  {m['metric']:<28} {fmt(m['baseline']):>12} "
-            f"{fmt(m['candidate']):>12} {delta_s:>14}  {m['status']}"
-        )
     per_rank = [
-        r for r in result.get("per_rank_step_time", [])
+        r
+        for r in result.get("per_rank_step_time", [])
         if r["baseline"] is not None or r["candidate"] is not None
     ]
     if len(per_rank) > 1:
@@ -237,13 +193,17 @@ def compare_files(
     path_a: str, path_b: str, fail_on_regression: bool = False
 ) -> int:
     try:
-        baseline = load_summary(path_a)
-        candidate = load_summary(path_b)
+        pair = compare_io.load_pair(path_a, path_b)
     except (OSError, ValueError) as exc:
         print(f"compare: {exc}", file=sys.stderr)
         return 1
-    result = compare_payloads(baseline, candidate)
-    print(render_compare(result))
+    result = compare_payloads(pair["baseline"], pair["candidate"])
+    result["notes"] = pair["notes"] + result.get("notes", [])
+    print(
+        render_compare(
+            result, pair["baseline_label"], pair["candidate_label"]
+        )
+    )
     if fail_on_regression and result["verdict"] == "REGRESSION":
         return 4  # CI perf gate
     return 0
