@@ -205,7 +205,8 @@ def _enrich_actions_from_code_manifest(payload: dict, session_dir: str) -> None:
     num_workers=0 (or unset), say so in the action — the most common cause,
     named concretely."""
     primary = payload.get("primary_diagnosis", {})
-    if primary.get("kind") not in ("INPUT_BOUND", "INPUT_STRAGGLER"):
+    kind = primary.get("kind")
+    if kind not in ("INPUT_BOUND", "INPUT_STRAGGLER", "H2D_BOUND"):
         return
     import json as _json
 
@@ -216,22 +217,37 @@ def _enrich_actions_from_code_manifest(payload: dict, session_dir: str) -> None:
             manifest = _json.load(f)
     except (OSError, ValueError):
         return
+
+    def _apply(hint: str) -> None:
+        primary["action"] = primary.get("action", "") + hint
+        st_diag = payload.get("step_time", {}).get("diagnosis")
+        if st_diag:
+            st_diag["action"] = st_diag.get("action", "") + hint
+
     for call in manifest.get("calls", []):
         if call.get("call") != "DataLoader":
             continue
-        workers = call.get("kwargs", {}).get("num_workers", 0)
-        if workers in (0, None, "<dynamic>") and workers != "<dynamic>":
-            hint = (
-                f" Your script constructs DataLoader with num_workers="
-                f"{workers} (line {call.get('line')}): loading happens on "
-                "the training thread — start with num_workers=8 and "
-                "pin_memory=True."
-            )
-            primary["action"] = primary.get("action", "") + hint
-            st_diag = payload.get("step_time", {}).get("diagnosis")
-            if st_diag:
-                st_diag["action"] = st_diag.get("action", "") + hint
-            break
+        kwargs = call.get("kwargs", {})
+        if kind in ("INPUT_BOUND", "INPUT_STRAGGLER"):
+            workers = kwargs.get("num_workers", 0)
+            if workers in (0, None):
+                _apply(
+                    f" Your script constructs DataLoader with num_workers="
+                    f"{workers} (line {call.get('line')}): loading happens on "
+                    "the training thread — start with num_workers=8 and "
+                    "pin_memory=True."
+                )
+                break
+        else:  # H2D_BOUND
+            pin = kwargs.get("pin_memory", False)
+            if pin in (False, None):
+                _apply(
+                    f" Your script constructs DataLoader without pin_memory "
+                    f"(line {call.get('line')}): page-locked staging enables "
+                    "async H2D — set pin_memory=True and pass "
+                    "non_blocking=True to .to(device)."
+                )
+                break
 
 
 def generate_summary(
