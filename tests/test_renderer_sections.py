@@ -355,3 +355,53 @@ def test_cli_driver_renders_sections(db_path):
     assert "INPUT STRAGGLER" in text or "STRAGGLER" in text
     assert "r2" in text  # straggler rank column present
     assert "mem alloc/reserved" in text
+
+
+# ---------------------------------------------------------------------------
+# HTML report depth
+# ---------------------------------------------------------------------------
+
+
+def test_html_per_rank_phase_bars_and_comm(db_path, tmp_path):
+    scenarios.input_straggler(ranks=4, steps=30).write(db_path)
+    _write_rank_stats_row(db_path, [
+        dict(rank=0, step=29, input_ms=4.0, forward_ms=30.0, backward_ms=235.0,
+             optimizer_ms=8.0, step_ms=280.0, ddp_comm_ms=185.0,
+             peak_alloc_bytes=0.0),
+        dict(rank=2, step=29, input_ms=184.0, forward_ms=30.0,
+             backward_ms=55.0, optimizer_ms=8.0, step_ms=280.0,
+             ddp_comm_ms=5.0, peak_alloc_bytes=0.0),
+    ])
+    from traceml_amd.reporting.final import FinalReportGenerator
+    from traceml_amd.reporting.html.document import render_html
+
+    payload = FinalReportGenerator(db_path).generate()
+    html = render_html(payload)
+    assert "Per-rank phase shares" in html
+    # 4 rank labels in the stacked bars
+    for rank in range(4):
+        assert f">r{rank}</text>" in html
+    assert "RCCL rank stats" in html
+    assert "gather latency" in html
+    assert "184.0" in html  # straggler input visible in the comm table
+
+
+def test_html_memory_trend_chart(db_path):
+    scenarios.healthy_ddp(ranks=1, steps=10).write(db_path)
+    scenarios.write_memory_rows(
+        db_path, {0: (40 << 30, 60 << 30), 1: (42 << 30, 60 << 30)},
+        capacity=288 << 30, steps=40, creep_bytes_per_step=32 << 20,
+    )
+    from traceml_amd.reporting.final import FinalReportGenerator
+    from traceml_amd.reporting.html.document import render_html
+
+    payload = FinalReportGenerator(db_path).generate()
+    evidence = payload["step_memory"]["evidence_extra"]
+    assert evidence["capacity_bytes"] == 288 << 30
+    assert evidence["trend"]["0"]["slope_bytes_per_step"] == pytest.approx(
+        32 << 20, rel=0.05
+    )
+    html = render_html(payload)
+    assert "Memory trend" in html
+    assert "capacity 288 GiB" in html
+    assert "MiB/step" in html  # slope annotation on the rising rank

@@ -71,6 +71,33 @@ def build(db_path: str) -> dict:
         if row is not None:
             row["identity"] = dict(s.identity)
 
+    # evidence for charts (HTML report / dashboard): capacity, trend slope
+    # and a downsampled per-rank allocated series
+    from traceml_amd.renderers.step_memory import _slope_bytes_per_step
+
+    trend = {}
+    capacity = None
+    for rank, s in sorted(series.items()):
+        if s.capacity is not None:
+            capacity = max(capacity or 0, s.capacity)
+        stride = max(1, len(s.steps) // 60)
+        points = [
+            [s.steps[i], s.peak_allocated[i]]
+            for i in range(0, len(s.steps), stride)
+            if s.peak_allocated[i] is not None
+        ]
+        if points:
+            trend[str(rank)] = {
+                "slope_bytes_per_step": _slope_bytes_per_step(
+                    s.steps, s.peak_allocated
+                ),
+                "series": points,
+            }
+    payload["evidence_extra"] = {
+        "capacity_bytes": capacity,
+        "trend": trend,
+    }
+
     diag = payload.get("diagnosis") or {}
     gib = 1 << 30
     lines = ["Step Memory"]

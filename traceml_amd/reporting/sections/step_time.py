@@ -81,7 +81,8 @@ def _latest_rank_stats(db_path: str) -> dict:
         conn = sqlite3.connect(f"file:{db_path}?mode=ro", uri=True)
         try:
             row = conn.execute(
-                "SELECT timestamp, world_size_gathered, ranks_json "
+                "SELECT timestamp, world_size_gathered, ranks_json, "
+                "gather_latency_ms, gather_latency_ms_mean "
                 "FROM rank_stats ORDER BY id DESC LIMIT 1"
             ).fetchone()
         finally:
@@ -98,6 +99,8 @@ def _latest_rank_stats(db_path: str) -> dict:
         "timestamp": row[0],
         "world_size": row[1],
         "ranks": ranks,
+        "gather_latency_ms": row[3],
+        "gather_latency_ms_mean": row[4],
     }
 
 
