@@ -26,8 +26,15 @@ def _target_device(args: tuple, kwargs: dict) -> Optional["object"]:
                     break
                 except (RuntimeError, ValueError):
                     continue
+            if isinstance(a, bool):
+                continue  # non_blocking/copy positionals
             if isinstance(a, int):
-                continue
+                # torch semantics: .to(0) targets cuda:0
+                try:
+                    device = torch.device(a)
+                    break
+                except (RuntimeError, ValueError):
+                    continue
             if isinstance(a, torch.Tensor):
                 device = a.device
                 break
