@@ -377,3 +377,72 @@ def test_yaml_precedence_corners(tmp_path, monkeypatch):
     s = resolve_config(cli_overrides={"interval": 1.0})
     assert s.interval == 1.0  # cli > env > yaml
     assert s.mode == "dashboard"  # env > yaml where no cli override
+
+
+# ---------------------------------------------------------------------------
+# allocator churn (allocator stats beyond peaks)
+# ---------------------------------------------------------------------------
+
+
+def test_allocator_churn_diagnosis(tmp_path):
+    """alloc_retries in the step-memory rows surface as an ALLOCATOR_CHURN
+    warning with the retry count and segment evidence."""
+    import sqlite3 as _sq
+
+    from traceml_amd.aggregator.writers import build_all_writers
+    from traceml_amd.diagnostics.step_memory.api import (
+        diagnose_step_memory,
+        load_memory_series,
+    )
+
+    db = str(tmp_path / "churn.sqlite")
+    conn = _sq.connect(db)
+    for w in build_all_writers():
+        w.init_schema(conn)
+    with conn:
+        for step in range(1, 31):
+            conn.execute(
+                "INSERT INTO step_memory_samples (global_rank, world_size,"
+                " timestamp, step, peak_allocated_bytes, peak_reserved_bytes,"
+                " device_capacity_bytes, device, active_peak_bytes,"
+                " alloc_retries, segments) VALUES (0,1,?,?,?,?,?,?,?,?,?)",
+                (time.time(), step, 10 << 30, 12 << 30, 288 << 30, "cuda:0",
+                 11 << 30, 1 if step % 10 == 0 else 0, 120),
+            )
+    conn.close()
+    series = load_memory_series(db)
+    assert series[0].alloc_retries_total == 3
+    assert series[0].segments_latest == 120
+    result = diagnose_step_memory(series)
+    churn = next(i for i in result.issues if i.kind == "ALLOCATOR_CHURN")
+    assert churn.severity == "warn"
+    assert "3" in churn.summary
+    assert churn.evidence["segments"] == 120
+
+
+def test_legacy_db_without_churn_columns_still_loads(tmp_path):
+    """Pre-churn-column DBs (round-1 artifacts) load via the legacy SELECT."""
+    import sqlite3 as _sq
+
+    from traceml_amd.diagnostics.step_memory.api import load_memory_series
+
+    db = str(tmp_path / "old.sqlite")
+    conn = _sq.connect(db)
+    conn.execute(
+        "CREATE TABLE step_memory_samples (id INTEGER PRIMARY KEY,"
+        " global_rank INTEGER, local_rank INTEGER, node_rank INTEGER,"
+        " hostname TEXT, world_size INTEGER, local_world_size INTEGER,"
+        " timestamp REAL, step INTEGER, peak_allocated_bytes INTEGER,"
+        " peak_reserved_bytes INTEGER, device_capacity_bytes INTEGER,"
+        " device TEXT)"
+    )
+    with conn:
+        conn.execute(
+            "INSERT INTO step_memory_samples (global_rank, step,"
+            " peak_allocated_bytes, peak_reserved_bytes,"
+            " device_capacity_bytes) VALUES (0, 1, 100, 200, 1000)"
+        )
+    conn.close()
+    series = load_memory_series(db)
+    assert series[0].peak_allocated == [100]
+    assert series[0].alloc_retries_total is None

@@ -109,6 +109,9 @@ class StepMemoryWriter(ProjectionWriter):
                 ("peak_reserved_bytes", "INTEGER"),
                 ("device_capacity_bytes", "INTEGER"),
                 ("device", "TEXT"),
+                ("active_peak_bytes", "INTEGER"),
+                ("alloc_retries", "INTEGER"),
+                ("segments", "INTEGER"),
             ],
         )
     }

@@ -30,6 +30,14 @@ class StepMemoryEvent:
     peak_reserved_bytes: Optional[int]
     device_capacity_bytes: Optional[int]
     device: Optional[str]
+    #: HIP caching-allocator churn stats (torch.cuda.memory_stats — the
+    #: native allocator on ROCm): per-step peak of ACTIVE bytes (allocated
+    #: minus freed-but-cached), the number of blocking cudaMalloc retries
+    #: this step (cache thrash: the allocator had to flush + re-reserve),
+    #: and the live segment count (fragmentation proxy).
+    active_peak_bytes: Optional[int] = None
+    alloc_retries: Optional[int] = None
+    segments: Optional[int] = None
 
 
 _queue_lock = threading.Lock()
@@ -63,6 +71,7 @@ class StepMemoryTracker:
         self._cuda = _cuda()
         self._device_index: Optional[int] = None
         self._capacity: Optional[int] = None
+        self._last_retries: Optional[int] = None
         if self._cuda is not None:
             try:
                 self._device_index = self._cuda.current_device()
@@ -83,6 +92,9 @@ class StepMemoryTracker:
         peak_alloc: Optional[int] = None
         peak_reserved: Optional[int] = None
         device: Optional[str] = None
+        active_peak: Optional[int] = None
+        retries_delta: Optional[int] = None
+        segments: Optional[int] = None
         if self._cuda is not None:
             try:
                 peak_alloc = int(self._cuda.max_memory_allocated(self._device_index))
@@ -90,6 +102,19 @@ class StepMemoryTracker:
                 device = f"cuda:{self._device_index}"
             except Exception:
                 peak_alloc = peak_reserved = None
+            # allocator-churn stats (one dict read; a few µs). Peaks reset
+            # with reset_peak_memory_stats; retries are cumulative -> delta.
+            try:
+                stats = self._cuda.memory_stats(self._device_index)
+                active_peak = stats.get("active_bytes.all.peak")
+                segments = stats.get("segment.all.current")
+                retries_total = stats.get("num_alloc_retries")
+                if retries_total is not None:
+                    if self._last_retries is not None:
+                        retries_delta = int(retries_total - self._last_retries)
+                    self._last_retries = int(retries_total)
+            except Exception:
+                pass
         event = StepMemoryEvent(
             step=step,
             timestamp=time.time(),
@@ -97,6 +122,9 @@ class StepMemoryTracker:
             peak_reserved_bytes=peak_reserved,
             device_capacity_bytes=self._capacity,
             device=device,
+            active_peak_bytes=active_peak,
+            alloc_retries=retries_delta,
+            segments=segments,
         )
         if peak_alloc is not None:
             global _last_peak_alloc

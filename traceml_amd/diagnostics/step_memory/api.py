@@ -32,6 +32,10 @@ class MemorySeries:
     peak_reserved: List[Optional[int]] = field(default_factory=list)
     capacity: Optional[int] = None
     identity: dict = field(default_factory=dict)
+    #: HIP caching-allocator churn: total blocking-malloc retries and the
+    #: latest live segment count over the window (None = not measured)
+    alloc_retries_total: Optional[int] = None
+    segments_latest: Optional[int] = None
 
 
 def load_memory_series(db_path: str) -> Dict[int, MemorySeries]:
@@ -47,12 +51,25 @@ def load_memory_series(db_path: str) -> Dict[int, MemorySeries]:
                 SELECT global_rank, local_rank, node_rank, hostname,
                        world_size, local_world_size, step,
                        peak_allocated_bytes, peak_reserved_bytes,
-                       device_capacity_bytes
+                       device_capacity_bytes, alloc_retries, segments
                 FROM step_memory_samples ORDER BY global_rank, step, id
                 """
             ).fetchall()
         except sqlite3.Error:
-            return {}
+            # pre-churn-column DB (additive schema evolution): legacy set
+            try:
+                rows = conn.execute(
+                    """
+                    SELECT global_rank, local_rank, node_rank, hostname,
+                           world_size, local_world_size, step,
+                           peak_allocated_bytes, peak_reserved_bytes,
+                           device_capacity_bytes, NULL AS alloc_retries,
+                           NULL AS segments
+                    FROM step_memory_samples ORDER BY global_rank, step, id
+                    """
+                ).fetchall()
+            except sqlite3.Error:
+                return {}
     finally:
         conn.close()
     series: Dict[int, MemorySeries] = {}
@@ -74,6 +91,12 @@ def load_memory_series(db_path: str) -> Dict[int, MemorySeries]:
         s.peak_reserved.append(r["peak_reserved_bytes"])
         if r["device_capacity_bytes"]:
             s.capacity = int(r["device_capacity_bytes"])
+        if r["alloc_retries"] is not None and r["alloc_retries"] > 0:
+            s.alloc_retries_total = (s.alloc_retries_total or 0) + int(
+                r["alloc_retries"]
+            )
+        if r["segments"] is not None:
+            s.segments_latest = int(r["segments"])
         s.identity = {
             "global_rank": rank,
             "local_rank": r["local_rank"],
@@ -256,6 +279,41 @@ def diagnose_step_memory(series: Dict[int, MemorySeries]) -> DiagnosticResult:
         creep = _creep_issue(s)
         if creep is not None:
             issues.append(creep)
+
+    # allocator churn: any blocking-malloc retry means the HIP caching
+    # allocator flushed its cache mid-step (fragmentation / oversubscribed
+    # reserve) — visible long before an OOM, invisible in peaks alone
+    for rank, s in sorted(measured.items()):
+        if s.alloc_retries_total:
+            issues.append(
+                DiagnosticIssue(
+                    kind="ALLOCATOR_CHURN",
+                    status="ALLOCATOR CHURN",
+                    severity="warn",
+                    summary=(
+                        f"Rank r{rank}: the HIP caching allocator hit "
+                        f"{s.alloc_retries_total} blocking-malloc "
+                        "retr{} during the window — cache flushes stall "
+                        "the stream and precede OOMs.".format(
+                            "y" if s.alloc_retries_total == 1 else "ies"
+                        )
+                    ),
+                    action=(
+                        "Reduce fragmentation: avoid many transient "
+                        "odd-sized allocations, or set "
+                        "PYTORCH_HIP_ALLOC_CONF=expandable_segments:True "
+                        "(288 GB HBM3E leaves headroom — churn is usually "
+                        "fragmentation, not capacity)."
+                    ),
+                    metric="alloc_retries",
+                    ranks=[rank],
+                    score=float(s.alloc_retries_total),
+                    evidence={
+                        "alloc_retries": s.alloc_retries_total,
+                        "segments": s.segments_latest,
+                    },
+                )
+            )
 
     if not issues:
         issues.append(

@@ -59,6 +59,10 @@ class StepMemorySampleRow:
     peak_reserved_bytes: Optional[int] = None
     device_capacity_bytes: Optional[int] = None
     device: Optional[str] = None
+    #: HIP caching-allocator churn stats (additive schema evolution)
+    active_peak_bytes: Optional[int] = None
+    alloc_retries: Optional[int] = None
+    segments: Optional[int] = None
 
     def to_wire(self) -> dict:
         return asdict(self)

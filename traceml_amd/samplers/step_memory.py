@@ -23,5 +23,8 @@ class StepMemorySampler(BaseSampler):
                     "peak_reserved_bytes": event.peak_reserved_bytes,
                     "device_capacity_bytes": event.device_capacity_bytes,
                     "device": event.device,
+                    "active_peak_bytes": event.active_peak_bytes,
+                    "alloc_retries": event.alloc_retries,
+                    "segments": event.segments,
                 },
             )
