@@ -128,7 +128,7 @@ class Workload:
     arm measures its real integration path.
     """
 
-    init_mode = "auto"
+    init_kwargs = dict(mode="auto")
     dtype = "bf16"
 
     def __init__(self, args, device, use_gpu):
@@ -286,7 +286,13 @@ class GPT2Workload(Workload):
     invoked directly — the integration code under test is identical.
     ``--creep-mb`` retains fresh GPU tensors each step (creep injector)."""
 
-    init_mode = "manual"  # Lightning owns phase timing (manual mode)
+    # Lightning policy (reference: integrations/lightning.py:108-125): the
+    # callback owns forward/backward/optimizer; the global DataLoader and
+    # Tensor.to patches supply input-wait and H2D.
+    init_kwargs = dict(
+        mode="custom", patch_dataloader=True, patch_h2d=True,
+        patch_forward=False, patch_backward=False,
+    )
 
     def build(self):
         from traceml_amd.models.gpt2 import GPT2
@@ -478,7 +484,7 @@ def main():
     handle = lifecycle.start_runtime(settings, fail_open=False,
                                      register_atexit=False)
     traceml_amd.init(
-        mode=work.init_mode, aggregator_port=settings.aggregator_port
+        **work.init_kwargs, aggregator_port=settings.aggregator_port
     )
     if world_size > 1:
         from traceml_amd.parallel.ddp_hook import attach_ddp_comm_timing

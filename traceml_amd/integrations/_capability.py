@@ -8,11 +8,14 @@ from __future__ import annotations
 import sys
 from typing import Iterable
 
-#: integration name -> step-time streams it promises to emit
+#: integration name -> step-time streams it promises to emit (reference
+#: REQUIRED_STEP_TIME registry, tests/integrations/test_telemetry_conformance
+#: .py:48-63: HF and Lightning both owe the dataloader-fetch stream)
 REQUIRED_STREAMS = {
-    "huggingface": ("forward_time", "backward_time", "optimizer_step", "step_time"),
-    "lightning": ("forward_time", "backward_time", "optimizer_step", "step_time",
-                  "h2d_time"),
+    "huggingface": ("dataloader_next", "forward_time", "backward_time",
+                    "optimizer_step", "step_time"),
+    "lightning": ("dataloader_next", "forward_time", "backward_time",
+                  "optimizer_step", "step_time", "h2d_time"),
     "accelerate": ("forward_time", "backward_time", "optimizer_step", "step_time"),
     "ray": ("forward_time", "backward_time", "step_time"),
 }
@@ -24,12 +27,22 @@ _STREAM_TO_PATCH = {
     "dataloader_next": "patch_dataloader",
 }
 
+#: streams an integration's CALLBACK produces itself (no global patch
+#: needed), so the patch-policy check must not flag them — Lightning owns
+#: forward/backward/optimizer timing in its hooks
+CALLBACK_OWNED = {
+    "lightning": {"forward_time", "backward_time", "optimizer_step"},
+}
+
 
 def warn_if_missing_streams(integration: str, config) -> list:
     """Return (and print) the owed streams the current config cannot emit."""
     owed: Iterable[str] = REQUIRED_STREAMS.get(integration, ())
+    callback_owned = CALLBACK_OWNED.get(integration, set())
     missing = []
     for stream in owed:
+        if stream in callback_owned:
+            continue  # produced by the integration's own hooks
         patch_field = _STREAM_TO_PATCH.get(stream)
         if patch_field is None:
             continue  # stream produced by the integration itself (hooks)

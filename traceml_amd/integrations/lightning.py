@@ -20,9 +20,18 @@ logger = logging.getLogger(__name__)
 
 
 def init(**kwargs):
+    """Selective mode (reference: integrations/lightning.py:108-125): the
+    callback owns forward/backward/optimizer timing, but the GLOBAL
+    DataLoader patch supplies the input-wait stream and the Tensor.to patch
+    supplies H2D — without them the summary reports INCOMPLETE DATA for
+    the dataloader signal."""
     import traceml_amd
 
-    kwargs.setdefault("mode", "manual")
+    kwargs.setdefault("mode", "custom")
+    kwargs.setdefault("patch_dataloader", True)
+    kwargs.setdefault("patch_h2d", True)
+    kwargs.setdefault("patch_forward", False)
+    kwargs.setdefault("patch_backward", False)
     config = traceml_amd.init(**kwargs)
     from traceml_amd.integrations._capability import warn_if_missing_streams
 
