@@ -145,6 +145,10 @@ def test_step_memory_watermarks(armed_auto_config):
     assert last.peak_allocated_bytes and last.peak_allocated_bytes > 0
     assert last.peak_reserved_bytes >= last.peak_allocated_bytes
     assert last.device_capacity_bytes and last.device_capacity_bytes > 100 * (1 << 30)
+    # HIP caching-allocator churn stats (torch.cuda.memory_stats on ROCm)
+    assert last.active_peak_bytes and last.active_peak_bytes > 0
+    assert last.segments and last.segments > 0
+    assert last.alloc_retries is None or last.alloc_retries >= 0
 
 
 @requires_gpu
