@@ -405,3 +405,36 @@ def test_html_memory_trend_chart(db_path):
     assert "Memory trend" in html
     assert "capacity 288 GiB" in html
     assert "MiB/step" in html  # slope annotation on the rising rank
+
+
+# ---------------------------------------------------------------------------
+# dashboard API contract (served route == section renderers)
+# ---------------------------------------------------------------------------
+
+
+def test_dashboard_api_contract(db_path):
+    from fastapi.testclient import TestClient
+
+    from traceml_amd.aggregator.display.dashboard import build_app
+
+    scenarios.input_straggler(ranks=4, steps=30).write(db_path)
+    state = {"db": None}
+    client = TestClient(build_app(lambda: state["db"]))
+
+    # no db yet -> 503 (aggregator not ready)
+    assert client.get("/api/live").status_code == 503
+
+    state["db"] = db_path
+    page = client.get("/")
+    assert page.status_code == 200
+    assert "traceml-amd dashboard" in page.text
+
+    live = client.get("/api/live")
+    assert live.status_code == 200
+    payload = live.json()
+    assert set(payload["sections"]) == {
+        "step_time", "step_memory", "system", "process", "comm",
+    }
+    st = payload["sections"]["step_time"]
+    assert st["diagnosis"]["kind"] == "INPUT_STRAGGLER"
+    assert st["ranks"] == ["0", "1", "2", "3"]

@@ -250,6 +250,31 @@ def _live_payload(db_path: str, session=None) -> dict:
     return live_view(db_path, session=session)
 
 
+def build_app(get_db_path, get_session=lambda: None):
+    """FastAPI app factory (separate from the uvicorn server so the route
+    contract is testable with a plain TestClient)."""
+    from fastapi import FastAPI
+    from fastapi.responses import HTMLResponse, JSONResponse
+
+    app = FastAPI()
+
+    @app.get("/")
+    def index():
+        return HTMLResponse(_PAGE)
+
+    @app.get("/api/live")
+    def live():
+        db_path = get_db_path()
+        if db_path is None:
+            return JSONResponse({}, status_code=503)
+        try:
+            return JSONResponse(_live_payload(db_path, get_session()))
+        except Exception as exc:
+            return JSONResponse({"error": repr(exc)}, status_code=500)
+
+    return app
+
+
 class DashboardDisplayDriver(DisplayDriver):
     def __init__(self, port: int = 8765) -> None:
         self.port = port
@@ -258,35 +283,23 @@ class DashboardDisplayDriver(DisplayDriver):
         self._db_path: Optional[str] = None
         self._session = None
 
+    def _get_session(self):
+        if self._session is None and self._db_path is not None:
+            from traceml_amd.steptime.pipeline import LiveStepTimeSession
+
+            self._session = LiveStepTimeSession(self._db_path)
+        return self._session
+
     def start(self) -> None:
         try:
             import uvicorn
-            from fastapi import FastAPI
-            from fastapi.responses import HTMLResponse, JSONResponse
+
+            app = build_app(lambda: self._db_path, self._get_session)
         except Exception:
             logger.warning(
                 "traceml_amd: fastapi/uvicorn unavailable, dashboard disabled"
             )
             return
-
-        app = FastAPI()
-
-        @app.get("/")
-        def index():
-            return HTMLResponse(_PAGE)
-
-        @app.get("/api/live")
-        def live():
-            if self._db_path is None:
-                return JSONResponse({}, status_code=503)
-            try:
-                if self._session is None:
-                    from traceml_amd.steptime.pipeline import LiveStepTimeSession
-
-                    self._session = LiveStepTimeSession(self._db_path)
-                return JSONResponse(_live_payload(self._db_path, self._session))
-            except Exception as exc:
-                return JSONResponse({"error": repr(exc)}, status_code=500)
 
         config = uvicorn.Config(
             app, host="0.0.0.0", port=self.port, log_level="error"

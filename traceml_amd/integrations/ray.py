@@ -61,7 +61,11 @@ def start_aggregator_actor(settings=None):
         def endpoint(self):
             import socket
 
-            return socket.gethostbyname(socket.gethostname()), self._aggregator.port
+            try:
+                host = socket.gethostbyname(socket.gethostname())
+            except OSError:
+                host = "127.0.0.1"  # unresolvable container hostname
+            return host, self._aggregator.port
 
         def stop(self):
             self._aggregator.stop()
