@@ -574,7 +574,7 @@ def main():
                     "resnet50": "auto patches",
                     "mlp": "auto patches",
                     "llama3": "HF TraceMLTrainerCallback bracket",
-                    "gpt2": "Lightning TraceMLCallback hooks (manual mode)",
+                    "gpt2": "Lightning TraceMLCallback hooks (selective patches)",
                 }[args.model],
             },
         }
