@@ -20,6 +20,9 @@ from dataclasses import dataclass
 from typing import Deque, List, Optional
 
 STEP_MEMORY_QUEUE_MAX = 4096
+#: allocator-churn stats (memory_stats dict) are read every Nth step —
+#: cumulative counters make the sparse delta lossless for retries
+CHURN_SAMPLE_EVERY = 16
 
 
 @dataclass
@@ -102,19 +105,25 @@ class StepMemoryTracker:
                 device = f"cuda:{self._device_index}"
             except Exception:
                 peak_alloc = peak_reserved = None
-            # allocator-churn stats (one dict read; a few µs). Peaks reset
-            # with reset_peak_memory_stats; retries are cumulative -> delta.
-            try:
-                stats = self._cuda.memory_stats(self._device_index)
-                active_peak = stats.get("active_bytes.all.peak")
-                segments = stats.get("segment.all.current")
-                retries_total = stats.get("num_alloc_retries")
-                if retries_total is not None:
-                    if self._last_retries is not None:
-                        retries_delta = int(retries_total - self._last_retries)
-                    self._last_retries = int(retries_total)
-            except Exception:
-                pass
+            # allocator-churn stats: torch.cuda.memory_stats builds a
+            # ~100-entry dict (~50 µs) so it is read every CHURN_SAMPLE_EVERY
+            # steps, not every step. Retries are cumulative, so the delta
+            # over the sparse read still captures EVERY retry in between;
+            # active-peak/segments are trend signals where sparse is enough.
+            if step % CHURN_SAMPLE_EVERY == 0 or self._last_retries is None:
+                try:
+                    stats = self._cuda.memory_stats(self._device_index)
+                    active_peak = stats.get("active_bytes.all.peak")
+                    segments = stats.get("segment.all.current")
+                    retries_total = stats.get("num_alloc_retries")
+                    if retries_total is not None:
+                        if self._last_retries is not None:
+                            retries_delta = int(
+                                retries_total - self._last_retries
+                            )
+                        self._last_retries = int(retries_total)
+                except Exception:
+                    pass
         event = StepMemoryEvent(
             step=step,
             timestamp=time.time(),
