@@ -58,8 +58,17 @@ class RingStampBackend:
         device = torch.cuda.current_device()
         self._ext.init(device, int(os.environ.get("TRACEML_AMD_RING_SLOTS", "65536")))
         self._torch = torch
+        self._device = device
+        # torch.cuda.current_stream() builds a Python Stream object per call
+        # (~5-10 µs); the private raw-stream hook returns the bare HIP
+        # stream handle (the same shortcut torch.compile's generated code
+        # uses). mark() runs ~12x per traced step — this is the bracket's
+        # single hottest line.
+        self._raw_stream = getattr(torch._C, "_cuda_getCurrentRawStream", None)
 
     def mark(self) -> int:
+        if self._raw_stream is not None:
+            return self._ext.ring_mark(self._raw_stream(self._device))
         stream = self._torch.cuda.current_stream().cuda_stream
         return self._ext.ring_mark(stream)
 
