@@ -334,3 +334,57 @@ def test_final_summary_smoke_ws8(tmp_path):
     )
     manifest = json.loads((session / "manifest.json").read_text())
     assert manifest["status"] == "completed"
+
+
+@pytest.mark.timeout(420)
+def test_multinode_two_launchers_one_host(tmp_path):
+    """Real multi-node path on localhost: TWO launcher invocations
+    (--nnodes 2, node-rank 0/1) whose torchrun agents rendezvous into one
+    ws=4 job. Only node 0 spawns the aggregator; node 1's ranks stream to
+    it over TCP. The summary must carry all 4 ranks with node_rank 0 and 1
+    (SURVEY: multi-node was config plumbing only, never executed)."""
+    script = tmp_path / "train_mn.py"
+    script.write_text(SCRIPT.replace("n >= 60", "n >= 12"))
+    agg_port = free_port()
+    master_port = free_port()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "90"
+    env["MASTER_ADDR"] = "127.0.0.1"
+
+    def launcher(node_rank):
+        return subprocess.Popen(
+            [
+                sys.executable, "-m", "traceml_amd", "run",
+                "--nnodes", "2", "--node-rank", str(node_rank),
+                "--nproc-per-node", "2",
+                "--master-addr", "127.0.0.1",
+                "--master-port", str(master_port),
+                "--run-name", "mn",
+                "--logs-dir", str(tmp_path / f"logs{node_rank}"),
+                "--aggregator-port", str(agg_port),
+                str(script),
+            ],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True, cwd=REPO_ROOT,
+        )
+
+    p0 = launcher(0)
+    p1 = launcher(1)
+    out0, err0 = p0.communicate(timeout=380)
+    out1, err1 = p1.communicate(timeout=60)
+    assert p0.returncode == 0, err0[-3000:]
+    assert p1.returncode == 0, err1[-3000:]
+
+    session = tmp_path / "logs0" / "mn"
+    payload = json.loads((session / "final_summary.json").read_text())
+    st = payload["step_time"]
+    assert sorted(st["metadata"]["global_ranks_seen"]) == [0, 1, 2, 3]
+    node_ranks = {
+        row["identity"]["node_rank"]
+        for row in st["groups"]["rows"].values()
+    }
+    assert node_ranks == {0, 1}
+    assert not (session / "finalization_warning.json").exists()
+    # node 1 never spawned an aggregator of its own
+    assert not (tmp_path / "logs1" / "mn" / "final_summary.json").exists()

@@ -191,7 +191,17 @@ def launch_process(
                     print(f.read())
             except OSError:
                 pass
-    if settings.mode == "summary" and not summary_ok and telemetry_status == "ok":
+    # The artifact gate applies only to the aggregator OWNER (node 0):
+    # non-owner nodes stream telemetry to node 0's aggregator and never
+    # produce a local final_summary.json — that is correct multi-node
+    # behavior, not a failure (reference: commands.py:566-579 runs the gate
+    # where the aggregator was spawned).
+    if (
+        settings.mode == "summary"
+        and agg_config.is_owner
+        and not summary_ok
+        and telemetry_status == "ok"
+    ):
         print(
             "[TraceML-AMD] ERROR: summary mode but no final_summary.json was "
             "produced",
