@@ -163,3 +163,40 @@ def test_ray_import_guard():
 
     with pytest.raises(ImportError):
         ray_integration.TraceMLTorchTrainer(lambda cfg: None)
+
+
+@pytest.mark.timeout(180)
+def test_accelerate_real_framework_conformance(armed_auto_config):
+    """REAL accelerate (installed in this image): Accelerator.prepare wraps
+    model/optimizer/dataloader; every owed stream must survive the wrappers
+    (reference: tests/integrations/test_accelerate.py runs the framework)."""
+    accelerate = pytest.importorskip("accelerate")
+    from torch.utils.data import DataLoader, TensorDataset
+
+    from traceml_amd.integrations import accelerate as acc
+
+    accelerator = accelerate.Accelerator(cpu=True)
+    model = nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    ds = TensorDataset(torch.randn(64, 16), torch.randint(0, 4, (64,)))
+    dl = DataLoader(ds, batch_size=8)
+    model, opt, dl = accelerator.prepare(model, opt, dl)
+
+    loss_fn = nn.CrossEntropyLoss()
+    steps = 0
+    for x, y in dl:
+        with acc.trace_step(model):
+            opt.zero_grad()
+            loss = loss_fn(model(x), y)
+            accelerator.backward(loss)
+            opt.step()
+        steps += 1
+    assert steps == 8
+
+    names = _collected_event_names()
+    for stream in REQUIRED_STREAMS["accelerate"]:
+        assert _STREAM_TO_EVENT[stream] in names, (
+            f"accelerate wrappers lost stream {stream}"
+        )
+    # the dataloader patch sees Accelerate's DataLoaderShard iterator too
+    assert event_names.DATALOADER in names
