@@ -200,3 +200,49 @@ def test_accelerate_real_framework_conformance(armed_auto_config):
         )
     # the dataloader patch sees Accelerate's DataLoaderShard iterator too
     assert event_names.DATALOADER in names
+
+
+class _FakeDSEngine(nn.Module):
+    """Minimal DeepSpeed model_engine surface the trace_step recipe touches:
+    ``.module`` unwrap, ``engine(x)`` forwarding, ``engine.backward(loss)``
+    and ``engine.step()`` driving the inner optimizer (no deepspeed in this
+    image; mirrors the reference's fake-engine recipe test)."""
+
+    def __init__(self, model, optimizer):
+        super().__init__()
+        self.module = model
+        self._optimizer = optimizer
+
+    def forward(self, x):
+        return self.module(x)
+
+    def backward(self, loss):
+        loss.backward()
+
+    def step(self):
+        self._optimizer.step()
+        self._optimizer.zero_grad()
+
+
+def test_deepspeed_engine_recipe(armed_auto_config):
+    """The documented DeepSpeed recipe — trace_step(engine) around
+    engine(x)/engine.backward/engine.step — emits forward/backward/
+    optimizer/step streams through the engine wrapper (.module unwrap)."""
+    from traceml_amd.sdk.instrumentation import trace_step
+
+    model = nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    engine = _FakeDSEngine(model, opt)
+    loss_fn = nn.CrossEntropyLoss()
+    for _ in range(4):
+        with trace_step(engine):
+            x = torch.randn(8, 16)
+            y = torch.randint(0, 4, (8,))
+            loss = loss_fn(engine(x), y)
+            engine.backward(loss)
+            engine.step()
+    names = _collected_event_names()
+    assert event_names.STEP_TIME in names
+    assert event_names.FORWARD in names, "engine .module unwrap lost forward"
+    assert event_names.BACKWARD in names
+    assert event_names.OPTIMIZER in names
