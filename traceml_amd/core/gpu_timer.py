@@ -18,8 +18,10 @@ two on-device clocks:
 
 On a machine without a GPU the backend is ``None`` and all timing falls back
 to the CPU wall clock (the analyzer's clock-selection rule handles this).
-On a GPU machine the native extension is REQUIRED: a missing/unloadable
-extension raises instead of silently degrading to a non-native path.
+On a GPU machine the native extension is REQUIRED: ``preflight_check()``
+(run once from ``init()``) raises loudly when it cannot load, BEFORE any
+training step; after that gate the hot path fails open to the CPU clock
+(telemetry must never crash training — reference architecture.md:54-59).
 Set ``TRACEML_AMD_GPU_TIMER=off`` to explicitly disable GPU timing, or
 ``=torch`` to use torch.cuda.Event (debug only).
 """
