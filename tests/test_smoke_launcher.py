@@ -388,3 +388,33 @@ def test_multinode_two_launchers_one_host(tmp_path):
     assert not (session / "finalization_warning.json").exists()
     # node 1 never spawned an aggregator of its own
     assert not (tmp_path / "logs1" / "mn" / "final_summary.json").exists()
+
+
+@pytest.mark.timeout(240)
+def test_launcher_ephemeral_aggregator_port(tmp_path):
+    """--aggregator-port 0: the aggregator binds an OS-assigned port,
+    publishes it to <session>/aggregator.port, and the launcher exports the
+    real port to the ranks — zero collision risk on a shared box."""
+    script = tmp_path / "train_eph.py"
+    script.write_text(SCRIPT.replace("n >= 60", "n >= 12"))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "60"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "eph",
+            "--aggregator-port", "0",
+            "--master-port", str(free_port()),
+            str(script),
+        ],
+        env=env, capture_output=True, text=True, timeout=220, cwd=REPO_ROOT,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    session = tmp_path / "logs" / "eph"
+    port_file = json.loads((session / "aggregator.port").read_text())
+    assert port_file["port"] > 0
+    payload = json.loads((session / "final_summary.json").read_text())
+    assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 12

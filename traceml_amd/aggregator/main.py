@@ -17,6 +17,12 @@ from traceml_amd.aggregator.aggregator import (
 from traceml_amd.runtime.settings import TraceMLSettings
 
 
+def port_file_path(session_dir: str) -> str:
+    import os
+
+    return os.path.join(session_dir, "aggregator.port")
+
+
 def main() -> int:
     settings = TraceMLSettings.from_env()
     aggregator = TraceMLAggregator(settings)
@@ -29,6 +35,18 @@ def main() -> int:
     signal.signal(signal.SIGINT, _handle)
 
     aggregator.start()
+    # Publish the BOUND port (which may be ephemeral when the launcher was
+    # given --aggregator-port 0): the launcher reads this file and exports
+    # the real port to the training ranks before torchrun spawns them.
+    try:
+        from traceml_amd.utils.atomic_io import atomic_write_json
+
+        atomic_write_json(
+            port_file_path(aggregator.session_dir),
+            {"port": aggregator.port, "bind": settings.aggregator_bind},
+        )
+    except OSError:
+        pass
     print(
         f"[TraceML-AMD] aggregator listening on "
         f"{settings.aggregator_bind}:{aggregator.port} "

@@ -38,6 +38,29 @@ def _executor_path() -> str:
     return os.path.abspath(executor.__file__)
 
 
+def _wait_for_port_file(sdir: str, proc, timeout: float):
+    """Poll <session>/aggregator.port for the ephemeral bound port."""
+    import json as _json
+    import time as _time
+
+    from traceml_amd.aggregator.main import port_file_path
+
+    path = port_file_path(sdir)
+    deadline = _time.time() + timeout
+    while _time.time() < deadline:
+        if proc is not None and proc.poll() is not None:
+            return None  # aggregator died before publishing
+        try:
+            with open(path, "r", encoding="utf-8") as f:
+                port = int(_json.load(f)["port"])
+            if port > 0:
+                return port
+        except (OSError, ValueError, KeyError, TypeError):
+            pass
+        _time.sleep(0.1)
+    return None
+
+
 def launch_process(
     script: str,
     script_args: List[str],
@@ -105,10 +128,44 @@ def launch_process(
 
     try:
         if agg_config.is_owner:
+            if settings.aggregator_port == 0:
+                # a stale port file from a reused session must not win the
+                # race against the new aggregator's publish
+                from traceml_amd.aggregator.main import port_file_path
+
+                try:
+                    os.remove(port_file_path(sdir))
+                except OSError:
+                    pass
             agg_proc, _ = spawn_process_group(
                 [sys.executable, "-m", "traceml_amd.aggregator.main"], env=env
             )
-            if not wait_for_tcp_listen(
+            if settings.aggregator_port == 0:
+                # Ephemeral port: the aggregator publishes its bound port to
+                # <session>/aggregator.port; export the real port to the
+                # ranks BEFORE torchrun spawns them. (Requires single-node:
+                # other nodes cannot read this file.)
+                if nnodes > 1:
+                    print(
+                        "[TraceML-AMD] --aggregator-port 0 requires a fixed "
+                        "port for multi-node runs (other nodes cannot "
+                        "discover it); continuing degraded",
+                        file=sys.stderr,
+                    )
+                    telemetry_status = "degraded"
+                else:
+                    discovered = _wait_for_port_file(sdir, agg_proc, 30.0)
+                    if discovered is None:
+                        print(
+                            "[TraceML-AMD] aggregator never published its "
+                            "port; continuing without telemetry",
+                            file=sys.stderr,
+                        )
+                        telemetry_status = "degraded"
+                    else:
+                        settings.aggregator_port = discovered
+                        env["TRACEML_AGGREGATOR_PORT"] = str(discovered)
+            if telemetry_status == "ok" and not wait_for_tcp_listen(
                 settings.aggregator_host,
                 settings.aggregator_port,
                 timeout=30.0,
