@@ -168,7 +168,10 @@ class RankStatsExchange:
             return
         with self._lock:
             self._poll_locked()
-            if step % self._every != 0:
+            # the FIRST flushed step is always a gate (short runs would
+            # otherwise never gather); ranks see the same first step value
+            # in lockstep DDP, so the decision stays identical everywhere
+            if self._last_gate_step is not None and step % self._every != 0:
                 return
             if step == self._last_gate_step:
                 return  # idempotent within one step value
