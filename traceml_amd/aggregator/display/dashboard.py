@@ -61,7 +61,8 @@ const PHASE_COLORS={input:'#e07b39',h2d:'#8e44ad',forward:'#2d7dd2',
   backward:'#1b998b',optimizer:'#97cc04',ddp_comm:'#d05ce3',residual:'#777'};
 const BAND={ok:'#1b998b',warn:'#f0ad4e',crit:'#d9534f',low:'#f0ad4e',
   moderate:'#dd6'};
-const esc=s=>String(s).replace(/</g,'&lt;');
+const esc=s=>String(s).replace(/&/g,'&amp;').replace(/</g,'&lt;')
+  .replace(/>/g,'&gt;').replace(/"/g,'&quot;');
 const fm=(v,d=1)=>v==null?'—':Number(v).toFixed(d);
 
 function meter(frac,bandName){
