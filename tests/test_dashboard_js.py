@@ -125,3 +125,18 @@ def test_dashboard_js_escapes_hostile_strings(tmp_path):
     assert "<script>alert(1)</script>" not in html
     assert "&lt;script&gt;" in html
     assert "<img src=x" not in html
+
+
+def test_dashboard_js_memory_sparkline(tmp_path):
+    """Creeping memory draws a rising (amber) sparkline in the card."""
+    db_path = str(tmp_path / "t.sqlite")
+    scenarios.healthy_ddp(ranks=1, steps=10).write(db_path)
+    scenarios.write_memory_rows(
+        db_path, {0: (10 << 30, 12 << 30)}, capacity=288 << 30,
+        steps=40, creep_bytes_per_step=128 << 20,
+    )
+    from traceml_amd.renderers import live_view
+
+    html = _render(live_view(db_path))
+    assert "polyline" in html
+    assert "#f0ad4e" in html  # rising series rendered in the warn color

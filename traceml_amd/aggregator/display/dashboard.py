@@ -133,6 +133,21 @@ function historyChart(history){
   return '<h2>Step-time history</h2>'+svg+'</svg>';
 }
 
+function memorySpark(spark){
+  if(!spark||spark.length<4) return '';
+  const W=170,H=26;
+  const xs=spark.map(p=>p[0]),ys=spark.map(p=>p[1]);
+  const x0=Math.min(...xs),x1=Math.max(...xs);
+  const y0=Math.min(...ys),y1=Math.max(...ys);
+  const span=Math.max(1,y1-y0);
+  const pts=spark.map(p=>
+    `${((p[0]-x0)/Math.max(1,x1-x0)*W).toFixed(1)},${(H-(p[1]-y0)/span*H).toFixed(1)}`
+  ).join(' ');
+  const rising=(y1-y0)/Math.max(1,y0)>0.02;
+  return `<svg width="${W}" height="${H}" xmlns="http://www.w3.org/2000/svg">`+
+    `<polyline points="${pts}" fill="none" stroke="${rising?'#f0ad4e':'#1b998b'}" stroke-width="1.2"/></svg>`;
+}
+
 function memorySection(sm){
   if(!sm||!sm.available||!sm.cards.length) return '';
   let html='<h2>Peak memory (HIP caching allocator)</h2><div class="cards">';
@@ -140,6 +155,7 @@ function memorySection(sm){
     const trend=c.trend_bytes_per_step;
     const trendTxt=trend==null?'':(trend>1024?
       `<div class="dim">trend +${(trend/1048576).toFixed(2)} MiB/step</div>`:'');
+    const spark=memorySpark(c.spark);
     const over=c.overhang_ratio!=null&&c.overhang_ratio>=2?
       `<div class="b-warn dim">reserved ${c.overhang_ratio.toFixed(1)}x allocated</div>`:'';
     html+=`<div class="card"><div class="t">rank ${c.rank}</div>
@@ -147,7 +163,7 @@ function memorySection(sm){
       <div class="dim">${c.peak_reserved_gib==null?'—':c.peak_reserved_gib+' GiB reserved of '+(c.capacity_gib||'?')+' GiB</div>'}
       ${meter(c.pressure_fraction,c.pressure_band||'ok')}
       <div class="dim">${c.pressure_fraction==null?'':(c.pressure_fraction*100).toFixed(0)+'% of capacity · '+c.steps_observed+' steps'}</div>
-      ${trendTxt}${over}</div>`;
+      ${spark}${trendTxt}${over}</div>`;
   }
   return html+'</div>';
 }
