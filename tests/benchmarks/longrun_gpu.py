@@ -1,0 +1,124 @@
+"""Production-soak: the FULL stack (launcher + aggregator + sampler + TCP +
+SQLite + dashboard-off) left on for tens of thousands of traced steps.
+
+Evidence for the "safe to leave on in production" claim beyond short
+benches: the profiler's own memory must be flat (bounded deques, ring
+recycling), no step-time batches dropped, and the whole window resolved.
+
+Run on a GPU box:  python tests/benchmarks/longrun_gpu.py [steps]
+Prints a PASS/FAIL summary of self-health checks.
+"""
+
+import json
+import os
+import sqlite3
+import subprocess
+import sys
+import time
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, REPO_ROOT)
+
+STEPS = int(sys.argv[1]) if len(sys.argv) > 1 else 20000
+
+TRAIN = f"""
+import torch, torch.nn.functional as F
+import traceml_amd
+from traceml_amd.models.resnet import resnet50
+
+traceml_amd.init()
+torch.backends.cudnn.benchmark = True
+model = resnet50().cuda().to(memory_format=torch.channels_last)
+opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+x = torch.randn(64, 3, 224, 224).contiguous(memory_format=torch.channels_last).pin_memory()
+y = torch.randint(0, 1000, (64,)).pin_memory()
+for step in range({STEPS}):
+    with traceml_amd.trace_step(model):
+        xs = x.to("cuda", non_blocking=True)
+        ys = y.to("cuda", non_blocking=True)
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", torch.bfloat16):
+            loss = F.cross_entropy(model(xs), ys)
+        loss.backward()
+        opt.step()
+print("train done", {STEPS})
+"""
+
+
+def main() -> int:
+    logs = os.path.join(REPO_ROOT, "gpurun_out", "longrun_logs")
+    script = os.path.join(logs, "train_long.py")
+    os.makedirs(logs, exist_ok=True)
+    with open(script, "w") as f:
+        f.write(TRAIN)
+    t0 = time.time()
+    proc = subprocess.run(
+        [sys.executable, "-m", "traceml_amd", "run",
+         "--logs-dir", logs, "--session-id", "longrun",
+         "--aggregator-port", "0", "--master-port", "29741", script],
+        cwd=REPO_ROOT, capture_output=True, text=True, timeout=1500,
+    )
+    wall = time.time() - t0
+    session = os.path.join(logs, "longrun")
+    checks = {}
+    ok = proc.returncode == 0
+    checks["launcher_rc"] = proc.returncode
+    if not ok:
+        print(proc.stderr[-3000:])
+
+    try:
+        with open(os.path.join(session, "final_summary.json")) as f:
+            payload = json.load(f)
+        st = payload["step_time"]["global"]["window"]
+        checks["steps_analyzed"] = st["steps_analyzed"]
+        checks["latest_step"] = payload["step_time"]["metadata"][
+            "training_latest_step"
+        ]
+        checks["primary"] = payload["primary_diagnosis"]["kind"]
+        avg = payload["step_time"]["global"]["average"]
+        checks["step_ms"] = round(avg["step_time_ms"], 3)
+        checks["gpu_clock"] = avg["step_time_gpu_ms"] is not None
+        # profiler self-RSS over the run: first vs last process sample
+        conn = sqlite3.connect(
+            os.path.join(session, "aggregator", "telemetry.sqlite")
+        )
+        first, last = conn.execute(
+            "SELECT (SELECT ram_bytes FROM process_samples ORDER BY id ASC "
+            "LIMIT 1), (SELECT ram_bytes FROM process_samples ORDER BY id "
+            "DESC LIMIT 1)"
+        ).fetchone()
+        n_mem = conn.execute(
+            "SELECT COUNT(*) FROM step_memory_samples"
+        ).fetchone()[0]
+        conn.close()
+        growth_mb = (last - first) / (1 << 20) if first and last else None
+        checks["rank_rss_growth_mb"] = None if growth_mb is None else round(
+            growth_mb, 1
+        )
+        checks["step_memory_rows"] = n_mem
+        # error log must not report dropped batches
+        errlog = os.path.join(session, "traceml_errors.log")
+        dropped = 0
+        if os.path.exists(errlog):
+            with open(errlog) as f:
+                dropped = f.read().count("dropped")
+        checks["dropped_mentions"] = dropped
+
+        ok = ok and checks["latest_step"] == STEPS
+        ok = ok and checks["gpu_clock"]
+        ok = ok and dropped == 0
+        # bounded telemetry: the training rank's RSS must stay flat-ish
+        # (the model/optimizer are constant; allow 512 MiB of allocator &
+        # cache settling over the whole run)
+        ok = ok and (growth_mb is None or growth_mb < 512)
+    except Exception as exc:
+        print("check failure:", repr(exc))
+        ok = False
+
+    checks["wall_s"] = round(wall, 1)
+    print("LONGRUN", "PASS" if ok else "FAIL", json.dumps(checks))
+    return 0 if ok else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
