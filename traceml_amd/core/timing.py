@@ -22,7 +22,7 @@ from dataclasses import dataclass, field
 from typing import Deque, List, Optional
 
 from traceml_amd.core import event_names, gpu_timer
-from traceml_amd.core.arming import is_tracing_armed, phase_flags
+from traceml_amd.core.arming import is_tracing_armed
 
 logger = logging.getLogger(__name__)
 

@@ -10,7 +10,6 @@ mode hard-fails without it) → manifest status.
 from __future__ import annotations
 
 import os
-import subprocess
 import sys
 import time
 from typing import List, Optional

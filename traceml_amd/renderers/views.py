@@ -2,7 +2,7 @@
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 #: display order + labels for the per-rank phase table
 STEP_TIME_TABLE_METRICS = (
