@@ -82,16 +82,25 @@ def main() -> int:
         conn = sqlite3.connect(
             os.path.join(session, "aggregator", "telemetry.sqlite")
         )
-        first, last = conn.execute(
-            "SELECT (SELECT ram_bytes FROM process_samples ORDER BY id ASC "
-            "LIMIT 1), (SELECT ram_bytes FROM process_samples ORDER BY id "
-            "DESC LIMIT 1)"
-        ).fetchone()
+        # steady-state RSS: compare the sample at the 25% mark (after CUDA
+        # context / MIOpen / pinned-buffer warmup, which legitimately adds
+        # >1 GB of host RSS) against the end of the run
+        rss = [
+            r[0]
+            for r in conn.execute(
+                "SELECT ram_bytes FROM process_samples ORDER BY id ASC"
+            )
+            if r[0]
+        ]
         n_mem = conn.execute(
             "SELECT COUNT(*) FROM step_memory_samples"
         ).fetchone()[0]
         conn.close()
-        growth_mb = (last - first) / (1 << 20) if first and last else None
+        growth_mb = (
+            (rss[-1] - rss[len(rss) // 4]) / (1 << 20)
+            if len(rss) >= 8
+            else None
+        )
         checks["rank_rss_growth_mb"] = None if growth_mb is None else round(
             growth_mb, 1
         )
@@ -107,10 +116,13 @@ def main() -> int:
         ok = ok and checks["latest_step"] == STEPS
         ok = ok and checks["gpu_clock"]
         ok = ok and dropped == 0
-        # bounded telemetry: the training rank's RSS must stay flat-ish
-        # (the model/optimizer are constant; allow 512 MiB of allocator &
-        # cache settling over the whole run)
-        ok = ok and (growth_mb is None or growth_mb < 512)
+        # no DataLoader in this loop: input is unmeasured (secondary
+        # incomplete-data note), but the verdict over the measured phases
+        # must still be the healthy COMPUTE_BOUND
+        ok = ok and checks["primary"] == "COMPUTE_BOUND"
+        # bounded telemetry: steady-state RSS must stay flat (deques are
+        # bounded, ring slots recycle); 256 MiB allows allocator jitter
+        ok = ok and (growth_mb is None or growth_mb < 256)
     except Exception as exc:
         print("check failure:", repr(exc))
         ok = False

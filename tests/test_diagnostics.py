@@ -247,3 +247,57 @@ def test_straggler_outranks_share_issue(db_path):
     kinds = [i.kind for i in result.diagnosis.issues]
     assert result.diagnosis.primary.kind == "INPUT_STRAGGLER"
     assert "INPUT_BOUND" in kinds  # the shared input-share issue still listed
+
+
+def test_missing_input_only_still_diagnoses_compute(tmp_path):
+    """A loop with NO DataLoader (pre-staged tensors — the production-soak
+    shape): input is never measured, but forward/backward/step are complete,
+    so the verdict must be the phase diagnosis over the measured signals
+    with incomplete-data demoted to a secondary note — not an
+    INSUFFICIENT_STEP_TIME_DATA primary (found by the 20k-step GPU soak)."""
+    from tests import scenarios
+    from traceml_amd.reporting.final import FinalReportGenerator
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    db = str(tmp_path / "noload.sqlite")
+
+    profile = scenarios.RankProfile(h2d_ms=0.6, forward_ms=30.0,
+                                    backward_ms=55.0, optimizer_ms=8.0)
+    original = profile.events
+
+    def events():
+        ev = original()
+        del ev[scenarios.event_names.DATALOADER]  # loop never fetches
+        return ev
+
+    profile.events = events  # type: ignore[method-assign]
+    scenarios.StepTimeScenario("noload", {0: profile}, steps=40).write(db)
+
+    result = StepTimePipeline(db, profile="summary").run()
+    assert result.diagnosis.primary.kind == "COMPUTE_BOUND"
+    kinds = [i.kind for i in result.diagnosis.issues]
+    assert "INCOMPLETE_DATA" in kinds  # caveat preserved, demoted
+    assert kinds[0] == "COMPUTE_BOUND"
+
+    payload = FinalReportGenerator(db).generate()
+    assert payload["primary_diagnosis"]["kind"] == "COMPUTE_BOUND"
+
+
+def test_missing_compute_signal_still_incomplete(tmp_path):
+    """Forward never measured -> incomplete-data stays THE verdict."""
+    from tests import scenarios
+    from traceml_amd.steptime.pipeline import StepTimePipeline
+
+    db = str(tmp_path / "nofwd.sqlite")
+    profile = scenarios.RankProfile()
+    original = profile.events
+
+    def events():
+        ev = original()
+        del ev[scenarios.event_names.FORWARD]
+        return ev
+
+    profile.events = events  # type: ignore[method-assign]
+    scenarios.StepTimeScenario("nofwd", {0: profile}, steps=40).write(db)
+    result = StepTimePipeline(db, profile="summary").run()
+    assert result.diagnosis.primary.kind == "INCOMPLETE_DATA"
