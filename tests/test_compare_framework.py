@@ -202,3 +202,52 @@ def test_compare_real_profile_artifacts(capsys):
         "step_time", "step_memory", "system", "process",
     }
     json.dumps(result)  # fully serializable
+
+
+# ---------------------------------------------------------------------------
+# property: compare never raises on arbitrarily mutated summaries
+# ---------------------------------------------------------------------------
+
+try:
+    from hypothesis import given, settings as hyp_settings, strategies as st
+
+    _HAVE_HYPOTHESIS = True
+except ImportError:  # pragma: no cover
+    _HAVE_HYPOTHESIS = False
+
+
+if _HAVE_HYPOTHESIS:
+    _scalars = st.one_of(
+        st.none(), st.booleans(), st.integers(-10**12, 10**12),
+        st.floats(allow_nan=False, allow_infinity=False), st.text(max_size=8),
+    )
+    _junk = st.recursive(
+        _scalars,
+        lambda children: st.one_of(
+            st.lists(children, max_size=3),
+            st.dictionaries(st.text(max_size=6), children, max_size=3),
+        ),
+        max_leaves=12,
+    )
+
+    @hyp_settings(max_examples=60, deadline=None)
+    @given(section=st.sampled_from(
+        ["step_time", "step_memory", "system", "process",
+         "primary_diagnosis", "meta"]),
+        junk=_junk)
+    def test_compare_never_raises_on_mutated_summary(section, junk, tmp_path_factory):
+        """Replacing any whole section with arbitrary JSON junk must never
+        crash compare_payloads — worst case is INCOMPARABLE."""
+        import copy
+
+        db = str(tmp_path_factory.mktemp("cmp") / "h.sqlite")
+        scenarios.healthy_ddp(ranks=1, steps=10).write(db)
+        base = FinalReportGenerator(db).generate()
+        mutated = copy.deepcopy(base)
+        mutated[section] = junk
+        result = compare_payloads(base, mutated)
+        assert result["verdict"] in (
+            "REGRESSION", "IMPROVEMENT", "NEUTRAL", "MIXED", "INCOMPARABLE",
+        )
+        json.dumps(result)
+        render_compare(result)

@@ -59,14 +59,16 @@ def compare_payloads(baseline: dict, candidate: dict) -> dict:
                 }
             )
 
+    def _primary_kind(payload: dict):
+        diag = payload.get("primary_diagnosis")
+        return diag.get("kind") if isinstance(diag, dict) else None
+
     st_diag = sections["step_time"].diagnosis
     return {
         "verdict": finding.verdict,
         "finding": finding.to_payload(),
-        "baseline_diagnosis": baseline.get("primary_diagnosis", {}).get("kind"),
-        "candidate_diagnosis": candidate.get("primary_diagnosis", {}).get(
-            "kind"
-        ),
+        "baseline_diagnosis": _primary_kind(baseline),
+        "candidate_diagnosis": _primary_kind(candidate),
         "diagnosis_transition": (
             st_diag.to_payload() if st_diag else None
         ),
