@@ -438,3 +438,27 @@ def test_dashboard_api_contract(db_path):
     st = payload["sections"]["step_time"]
     assert st["diagnosis"]["kind"] == "INPUT_STRAGGLER"
     assert st["ranks"] == ["0", "1", "2", "3"]
+
+
+def test_html_and_text_render_never_raise_on_mutated_payload(db_path):
+    """Every top-level section replaced by junk: render_html and the
+    verdict text builder must degrade, not raise (a partially-written
+    summary from a crashed run is re-read by `traceml-amd view`)."""
+    scenarios.healthy_ddp(ranks=2, steps=10).write(db_path)
+    from traceml_amd.reporting.final import FinalReportGenerator, build_verdict_text
+    from traceml_amd.reporting.html.document import render_html
+
+    base = FinalReportGenerator(db_path).generate()
+    junk_values = (None, 1, "x", [], {}, {"groups": None},
+                   {"global": {"average": None}},
+                   {"evidence_extra": {"trend": {"0": None}}})
+    import copy
+
+    for section in ("step_time", "step_memory", "system", "process",
+                    "primary_diagnosis", "meta"):
+        for junk in junk_values:
+            payload = copy.deepcopy(base)
+            payload[section] = junk
+            html = render_html(payload)
+            assert "<html" in html
+            build_verdict_text(payload)

@@ -113,8 +113,12 @@ class FinalReportGenerator:
         return payload
 
 
+def _junk_tolerant_dict(value) -> dict:
+    return value if isinstance(value, dict) else {}
+
+
 def build_verdict_text(payload: dict) -> str:
-    primary = payload.get("primary_diagnosis", {})
+    primary = _junk_tolerant_dict(payload.get("primary_diagnosis"))
     lines = [
         "TraceML-AMD Verdict: "
         + str(primary.get("status", "UNKNOWN"))
@@ -127,7 +131,9 @@ def build_verdict_text(payload: dict) -> str:
     lines.append("")
     lines.append("Sections:")
     for name in ("step_time", "step_memory", "system", "process"):
-        diag = payload.get(name, {}).get("diagnosis", {})
+        diag = _junk_tolerant_dict(
+            _junk_tolerant_dict(payload.get(name)).get("diagnosis")
+        )
         if diag:
             lines.append(
                 f"  {name:<12} {diag.get('status', ''):<28} "
@@ -139,7 +145,7 @@ def build_verdict_text(payload: dict) -> str:
         lines.append(rank_table)
         lines.append("")
     for name in ("step_time", "step_memory", "system", "process"):
-        card = payload.get(name, {}).get("card")
+        card = _junk_tolerant_dict(payload.get(name)).get("card")
         if card:
             lines.append(card)
             lines.append("")
@@ -148,7 +154,11 @@ def build_verdict_text(payload: dict) -> str:
 
 def _rank_evidence_table(payload: dict) -> str:
     """Compact per-rank step-time evidence (multi-rank runs only)."""
-    rows = payload.get("step_time", {}).get("groups", {}).get("rows", {})
+    rows = _junk_tolerant_dict(
+        _junk_tolerant_dict(
+            _junk_tolerant_dict(payload.get("step_time")).get("groups")
+        ).get("rows")
+    )
     if len(rows) < 2:
         return ""
     metrics = [

@@ -28,6 +28,12 @@ def _esc(value) -> str:
     return _html.escape(str(value if value is not None else ""))
 
 
+def _dict(value) -> dict:
+    """Junk-tolerant coercion: view/compare re-read summaries written by
+    crashed runs, so every section access assumes nothing."""
+    return value if isinstance(value, dict) else {}
+
+
 def _phase_bar_svg(shares: dict) -> str:
     """Horizontal stacked bar of phase shares of step time."""
     palette = {
@@ -80,7 +86,7 @@ def _per_rank_phase_bars(step_time_section: dict) -> str:
     """One stacked phase-share bar per rank: widths are
     each phase's share of THAT rank's step time, so a straggler's anomalous
     phase is visible at a glance across ranks."""
-    rows = step_time_section.get("groups", {}).get("rows", {})
+    rows = _dict(_dict(step_time_section.get("groups")).get("rows"))
     if len(rows) < 1:
         return ""
     total_w, bar_h, gap = 600, 18, 6
@@ -88,8 +94,10 @@ def _per_rank_phase_bars(step_time_section: dict) -> str:
     parts: List[str] = []
     y = 0
     for key in ranks:
-        metrics = rows[key].get("metrics", {})
+        metrics = _dict(_dict(rows.get(key)).get("metrics"))
         step = metrics.get("step_time_ms")
+        if not isinstance(step, (int, float)):
+            step = None
         if not step:
             continue
         parts.append(
@@ -136,10 +144,16 @@ _RANK_COLORS = ("#2d7dd2", "#1b998b", "#e07b39", "#d05ce3",
 def _memory_trend_svg(step_memory_section: dict) -> str:
     """Per-rank peak-allocated line chart over steps with the HBM capacity
     line — memory creep is visible as a rising slope (VERDICT r01 #9)."""
-    evidence = step_memory_section.get("evidence_extra") or {}
-    trend = evidence.get("trend") or {}
+    evidence = _dict(step_memory_section.get("evidence_extra"))
+    trend = _dict(evidence.get("trend"))
     series_by_rank = {
-        rank: t.get("series") or [] for rank, t in trend.items()
+        rank: (_dict(t).get("series") or []) for rank, t in trend.items()
+    }
+    series_by_rank = {
+        rank: [p for p in series if isinstance(p, (list, tuple))
+               and len(p) == 2 and all(isinstance(v, (int, float)) for v in p)]
+        for rank, series in series_by_rank.items()
+        if isinstance(series, list)
     }
     all_points = [p for s in series_by_rank.values() for p in s]
     if len(all_points) < 4:
@@ -147,6 +161,8 @@ def _memory_trend_svg(step_memory_section: dict) -> str:
     xs = [p[0] for p in all_points]
     ys = [p[1] for p in all_points]
     capacity = evidence.get("capacity_bytes")
+    if not isinstance(capacity, (int, float)):
+        capacity = None
     x0, x1 = min(xs), max(xs)
     y_top = max(max(ys) * 1.1, (capacity or 0) * 0.25)
     W, H = 600, 120
@@ -175,7 +191,7 @@ def _memory_trend_svg(step_memory_section: dict) -> str:
             f'<polyline points="{points}" fill="none" stroke="{color}" '
             'stroke-width="1.5"/>'
         )
-        slope = (trend.get(rank) or {}).get("slope_bytes_per_step")
+        slope = _dict(trend.get(rank)).get("slope_bytes_per_step")
         label = f"r{rank}"
         if isinstance(slope, (int, float)) and slope > 1024:
             label += f" (+{slope / (1 << 20):.2f} MiB/step)"
@@ -201,7 +217,7 @@ def _memory_trend_svg(step_memory_section: dict) -> str:
 def _comm_section(step_time_section: dict) -> str:
     """RCCL/xGMI rank-stats table from the step-time evidence (the new
     MI355X comm plane; VERDICT r01 #9)."""
-    comm = (step_time_section.get("evidence_extra") or {}).get(
+    comm = _dict(step_time_section.get("evidence_extra")).get(
         "rccl_rank_stats"
     )
     if not isinstance(comm, dict):
@@ -223,6 +239,9 @@ def _comm_section(step_time_section: dict) -> str:
     )
     body = []
     for r in ranks:
+        if not isinstance(r, dict):
+            continue
+
         def cell(key):
             v = r.get(key)
             return f"{v:.1f}" if isinstance(v, (int, float)) else "—"
@@ -239,14 +258,14 @@ def _comm_section(step_time_section: dict) -> str:
 
 
 def _rank_table(section: dict) -> str:
-    rows = section.get("groups", {}).get("rows", {})
-    metrics = section.get("metadata", {}).get("section_metric_names", [])
+    rows = _dict(_dict(section.get("groups")).get("rows"))
+    metrics = _dict(section.get("metadata")).get("section_metric_names") or []
     if not rows or not metrics:
         return ""
     header = "<tr><th>rank</th>" + "".join(f"<th>{_esc(m)}</th>" for m in metrics) + "</tr>"
     body = []
     for key in sorted(rows, key=lambda k: (len(k), k)):
-        cells = rows[key].get("metrics", {})
+        cells = _dict(_dict(rows.get(key)).get("metrics"))
         body.append(
             f"<tr><td>{_esc(key)}</td>"
             + "".join(
@@ -261,7 +280,7 @@ def _rank_table(section: dict) -> str:
 
 
 def render_html(payload: dict) -> str:
-    primary = payload.get("primary_diagnosis", {})
+    primary = _dict(payload.get("primary_diagnosis"))
     severity = primary.get("severity", "info")
     color = _SEVERITY_COLORS.get(severity, "#5bc0de")
     parts = [
@@ -277,21 +296,21 @@ def render_html(payload: dict) -> str:
     if action:
         parts.append(f"<p><b>Next:</b> {_esc(action)}</p>")
 
-    shares = (
-        payload.get("step_time", {}).get("evidence_extra", {}).get("shares") or {}
+    shares = _dict(
+        _dict(_dict(payload.get("step_time")).get("evidence_extra")).get("shares")
     )
     svg = _phase_bar_svg(shares)
     if svg:
         parts.append("<h2>Step-time phase breakdown</h2>" + svg)
-    parts.append(_per_rank_phase_bars(payload.get("step_time", {}) or {}))
-    parts.append(_memory_trend_svg(payload.get("step_memory", {}) or {}))
-    parts.append(_comm_section(payload.get("step_time", {}) or {}))
+    parts.append(_per_rank_phase_bars(_dict(payload.get("step_time"))))
+    parts.append(_memory_trend_svg(_dict(payload.get("step_memory"))))
+    parts.append(_comm_section(_dict(payload.get("step_time"))))
 
     for name in ("step_time", "step_memory", "system", "process"):
-        section = payload.get(name)
+        section = _dict(payload.get(name))
         if not section:
             continue
-        diag = section.get("diagnosis", {})
+        diag = _dict(section.get("diagnosis"))
         sev = diag.get("severity", "info")
         parts.append(
             f"<h2>{_esc(name)}<span class='badge' "
