@@ -116,3 +116,41 @@ def test_event_name_vocabulary_is_closed():
     assert set(timing._SUMMARY_KEYS) <= set(event_names.ALL_EVENT_NAMES)
     for name in event_names.ALL_EVENT_NAMES:
         assert name.startswith(event_names.PREFIX)
+
+
+def test_step_memory_msgpack_roundtrip_includes_churn_fields(tmp_path):
+    """The per-rank msgpack backups (read by `traceml-amd inspect`) carry
+    the allocator-churn columns end to end."""
+    from traceml_amd.database.database import Database
+    from traceml_amd.database.writer import DatabaseWriter, read_msgpack_table
+
+    db = Database()
+    db.add_record(
+        "step_memory_samples",
+        {"timestamp": 1.0, "step": 16, "peak_allocated_bytes": 100,
+         "peak_reserved_bytes": 200, "device_capacity_bytes": 1000,
+         "device": "cuda:0", "active_peak_bytes": 90, "alloc_retries": 1,
+         "segments": 7},
+    )
+    writer = DatabaseWriter("step_memory", db, str(tmp_path / "data"))
+    writer.flush()
+    writer.close()
+    path = next((tmp_path / "data").rglob("*.msgpack"))
+    rows = read_msgpack_table(str(path))
+    assert rows[0]["alloc_retries"] == 1
+    assert rows[0]["segments"] == 7
+    assert rows[0]["active_peak_bytes"] == 90
+
+
+def test_gpu_timer_env_off_disables_backend(monkeypatch):
+    """TRACEML_AMD_GPU_TIMER=off: no backend even where one would load,
+    and the preflight becomes a no-op (documented escape hatch)."""
+    from traceml_amd.core import gpu_timer
+
+    monkeypatch.setenv("TRACEML_AMD_GPU_TIMER", "off")
+    gpu_timer.reset_backend_for_tests()
+    try:
+        assert gpu_timer.get_backend() is None
+        gpu_timer.preflight_check()  # must not raise
+    finally:
+        gpu_timer.reset_backend_for_tests()
