@@ -507,9 +507,16 @@ def main():
     rounds = int(max_over_ranks(float(rounds), world_size, use_gpu))
 
     offs, ons = [], []
-    for _ in range(rounds):
-        ons.append(run_phase(work, model, args, world_size, use_gpu, traced=True))
-        offs.append(run_phase(work, model, args, world_size, use_gpu, traced=False))
+    # ABBA ordering: monotone clock drift (DVFS/thermal ramp over the run)
+    # would systematically favor whichever arm always runs second; flipping
+    # the order each round cancels first-order drift out of the medians.
+    for i in range(rounds):
+        order = (True, False) if i % 2 == 0 else (False, True)
+        for traced in order:
+            (ons if traced else offs).append(
+                run_phase(work, model, args, world_size, use_gpu,
+                          traced=traced)
+            )
     t_off_med = statistics.median(offs)
     t_on_med = statistics.median(ons)
     #: same-arm spread = measurement noise floor for this config
