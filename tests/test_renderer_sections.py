@@ -462,3 +462,39 @@ def test_html_and_text_render_never_raise_on_mutated_payload(db_path):
             html = render_html(payload)
             assert "<html" in html
             build_verdict_text(payload)
+
+
+def test_cross_surface_consistency_same_scenario(db_path):
+    """SURVEY §4's core contract: ONE scenario must produce the SAME
+    diagnosis and rank set on every surface — live payload, CLI panel,
+    final summary and the verdict text."""
+    scenarios.compute_straggler(ranks=4, steps=30).write(db_path)
+
+    from rich.console import Console
+
+    from traceml_amd.aggregator.display.cli import CLIDisplayDriver
+    from traceml_amd.renderers import live_view
+    from traceml_amd.reporting.final import FinalReportGenerator
+
+    live = live_view(db_path)
+    live_kind = live["sections"]["step_time"]["diagnosis"]["kind"]
+    live_ranks = live["sections"]["step_time"]["ranks"]
+
+    summary = FinalReportGenerator(db_path).generate()
+    summary_kind = summary["step_time"]["diagnosis"]["kind"]
+    summary_ranks = sorted(summary["step_time"]["groups"]["rows"])
+
+    console = Console(record=True, width=140)
+    console.print(CLIDisplayDriver()._build(db_path))
+    cli_text = console.export_text()
+
+    assert live_kind == summary_kind == "COMPUTE_STRAGGLER"
+    assert live_ranks == summary_ranks == ["0", "1", "2", "3"]
+    status = summary["step_time"]["diagnosis"]["status"]
+    assert status in cli_text
+    assert status in summary["text"]
+    # the culprit rank named by the diagnosis appears on every surface
+    culprit = summary["step_time"]["diagnosis"].get("ranks") or []
+    assert culprit, "diagnosis lost its culprit rank"
+    assert f"r{culprit[0]}" in cli_text
+    assert f"r{culprit[0]}" in summary["text"]
