@@ -99,31 +99,33 @@ class StepMemoryTracker:
         retries_delta: Optional[int] = None
         segments: Optional[int] = None
         if self._cuda is not None:
+            # ONE stats read per step: torch.cuda.max_memory_allocated and
+            # max_memory_reserved each build the full ~100-entry allocator
+            # stats dict internally (~40 µs apiece on a busy allocator), so
+            # both peaks are read from a single memory_stats() call — and
+            # the churn fields reuse the SAME dict on their sampled steps.
             try:
-                peak_alloc = int(self._cuda.max_memory_allocated(self._device_index))
-                peak_reserved = int(self._cuda.max_memory_reserved(self._device_index))
+                stats = self._cuda.memory_stats(self._device_index)
+                peak_alloc = int(stats.get("allocated_bytes.all.peak", 0))
+                peak_reserved = int(stats.get("reserved_bytes.all.peak", 0))
                 device = f"cuda:{self._device_index}"
             except Exception:
+                stats = None
                 peak_alloc = peak_reserved = None
-            # allocator-churn stats: torch.cuda.memory_stats builds a
-            # ~100-entry dict (~50 µs) so it is read every CHURN_SAMPLE_EVERY
-            # steps, not every step. Retries are cumulative, so the delta
-            # over the sparse read still captures EVERY retry in between;
-            # active-peak/segments are trend signals where sparse is enough.
-            if step % CHURN_SAMPLE_EVERY == 0 or self._last_retries is None:
-                try:
-                    stats = self._cuda.memory_stats(self._device_index)
-                    active_peak = stats.get("active_bytes.all.peak")
-                    segments = stats.get("segment.all.current")
-                    retries_total = stats.get("num_alloc_retries")
-                    if retries_total is not None:
-                        if self._last_retries is not None:
-                            retries_delta = int(
-                                retries_total - self._last_retries
-                            )
-                        self._last_retries = int(retries_total)
-                except Exception:
-                    pass
+            # Churn stats every CHURN_SAMPLE_EVERY steps: retries are
+            # cumulative, so the delta over the sparse read still captures
+            # EVERY retry in between; active-peak/segments are trend
+            # signals where sparse is enough.
+            if stats is not None and (
+                step % CHURN_SAMPLE_EVERY == 0 or self._last_retries is None
+            ):
+                active_peak = stats.get("active_bytes.all.peak")
+                segments = stats.get("segment.all.current")
+                retries_total = stats.get("num_alloc_retries")
+                if retries_total is not None:
+                    if self._last_retries is not None:
+                        retries_delta = int(retries_total - self._last_retries)
+                    self._last_retries = int(retries_total)
         event = StepMemoryEvent(
             step=step,
             timestamp=time.time(),
