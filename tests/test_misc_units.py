@@ -154,3 +154,84 @@ def test_gpu_timer_env_off_disables_backend(monkeypatch):
         gpu_timer.preflight_check()  # must not raise
     finally:
         gpu_timer.reset_backend_for_tests()
+
+
+def test_codec_json_fallback(monkeypatch):
+    """Without msgpack the codec transparently speaks JSON — both ends must
+    agree via the same module flags (reference: msgpack_codec.py:60-80)."""
+    from traceml_amd.transport import codec
+
+    payload = {"a": [1, 2.5, "x"], "meta": {"rank": 0}}
+    monkeypatch.setattr(codec, "_HAVE_MSGPACK", False)
+    data = codec.encode(payload)
+    assert data.startswith(b"{")  # JSON bytes
+    assert codec.decode(data) == payload
+    batch = codec.encode_batch([payload, payload])
+    assert codec.decode(batch) == [payload, payload]
+
+
+def test_codec_numpy_scalars_on_the_wire():
+    import numpy as np
+
+    from traceml_amd.transport import codec
+
+    decoded = codec.decode(codec.encode({"v": np.float64(1.5),
+                                         "n": np.int64(7)}))
+    assert decoded["v"] == 1.5 and decoded["n"] == 7
+
+
+def test_atomic_write_survives_interrupted_replacement(tmp_path, monkeypatch):
+    """A crash between tmp-write and rename must leave the OLD file intact
+    (the summary artifact is never half-written)."""
+    import os
+
+    from traceml_amd.utils import atomic_io
+
+    target = tmp_path / "final_summary.json"
+    atomic_io.atomic_write_json(str(target), {"v": 1})
+
+    real_replace = os.replace
+
+    def crashing_replace(src, dst):
+        raise OSError("simulated crash mid-replace")
+
+    monkeypatch.setattr(os, "replace", crashing_replace)
+    try:
+        atomic_io.atomic_write_json(str(target), {"v": 2})
+    except OSError:
+        pass
+    monkeypatch.setattr(os, "replace", real_replace)
+    import json
+
+    assert json.loads(target.read_text()) == {"v": 1}  # old content intact
+
+
+def test_exporter_final_drain_budget(monkeypatch):
+    """stop() flushes what it can inside the drain budget and returns —
+    a dead aggregator must not hang teardown (reference exporter:160-180)."""
+    import time as _time
+
+    from traceml_amd.runtime.exporter import TelemetryExporter
+
+    class _SlowClient:
+        def __init__(self):
+            self.sent = []
+
+        def send_batch(self, payloads):
+            _time.sleep(0.05)
+            self.sent.append(payloads)
+            return True
+
+        def close(self):
+            pass
+
+    client = _SlowClient()
+    exporter = TelemetryExporter(client)
+    exporter.start()
+    for i in range(50):
+        exporter.send_batch([{"i": i}])
+    start = _time.time()
+    exporter.stop()
+    elapsed = _time.time() - start
+    assert elapsed < 10.0  # bounded, not 50*0.05 + unbounded wait
+    assert client.sent  # at least part of the backlog flushed
