@@ -446,3 +446,66 @@ def test_legacy_db_without_churn_columns_still_loads(tmp_path):
     series = load_memory_series(db)
     assert series[0].peak_allocated == [100]
     assert series[0].alloc_retries_total is None
+
+
+@pytest.mark.timeout(120)
+def test_concurrent_clients_hammer_aggregator(tmp_path):
+    """8 client threads × 50 batches each, concurrent force_flush ticks:
+    every surviving row lands exactly once, no deadlock, queue counters
+    consistent (wire robustness under parallel load)."""
+    agg = _aggregator(tmp_path, "hammer", finalize_timeout=10.0)
+    port = agg.port
+    n_clients, n_batches = 8, 50
+    errors = []
+
+    def client_thread(rank):
+        try:
+            c = TCPClient("127.0.0.1", port)
+            for step in range(1, n_batches + 1):
+                c.send_batch([_step_envelope(rank, step)])
+            c.send_batch([build_rank_finished(_meta(rank))])
+            c.close()
+        except Exception as exc:  # must never happen: client is best-effort
+            errors.append(repr(exc))
+
+    threads = [
+        threading.Thread(target=client_thread, args=(r,))
+        for r in range(n_clients)
+    ]
+    flushers = [
+        threading.Thread(target=lambda: [agg.sqlite.force_flush(2.0)
+                                         for _ in range(5)])
+        for _ in range(2)
+    ]
+    for t in threads + flushers:
+        t.start()
+    for t in threads + flushers:
+        t.join(timeout=30)
+    assert not errors, errors
+    deadline = time.time() + 20
+    expected = n_clients * n_batches
+    count = 0
+    while time.time() < deadline:
+        agg.sqlite.force_flush(2.0)
+        conn = sqlite3.connect(agg.db_path)
+        try:
+            count = conn.execute(
+                "SELECT COUNT(*) FROM step_time_samples"
+            ).fetchone()[0]
+        finally:
+            conn.close()
+        if count >= expected:
+            break
+        time.sleep(0.2)
+    agg.stop()
+    assert count == expected, f"{count} != {expected} (lost or duplicated rows)"
+    # no duplicates per (rank, step)
+    conn = sqlite3.connect(agg.db_path)
+    try:
+        dupes = conn.execute(
+            "SELECT global_rank, step, COUNT(*) c FROM step_time_samples"
+            " GROUP BY global_rank, step HAVING c > 1"
+        ).fetchall()
+    finally:
+        conn.close()
+    assert not dupes, dupes[:5]
