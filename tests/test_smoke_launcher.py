@@ -418,3 +418,53 @@ def test_launcher_ephemeral_aggregator_port(tmp_path):
     assert port_file["port"] > 0
     payload = json.loads((session / "final_summary.json").read_text())
     assert payload["step_time"]["global"]["window"]["steps_analyzed"] == 12
+
+
+@pytest.mark.timeout(300)
+def test_ctrl_c_mid_run_still_produces_summary(tmp_path):
+    """SIGINT on the launcher mid-training: the signal is forwarded to the
+    torchrun and aggregator process groups, the aggregator finalizes
+    inside its budget, and the partial run still yields a usable
+    final_summary.json (the everyday interrupted-run workflow)."""
+    import signal as _signal
+    import time as _time
+
+    script = tmp_path / "train_forever.py"
+    script.write_text(SCRIPT.replace("n >= 60", "n >= 100000")
+                      .replace("for epoch in range(2):",
+                               "for epoch in range(10000):"))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["TRACEML_FINALIZE_TIMEOUT"] = "45"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "traceml_amd", "run",
+            "--logs-dir", str(tmp_path / "logs"),
+            "--session-id", "interrupted",
+            "--aggregator-port", str(free_port()),
+            "--master-port", str(free_port()),
+            str(script),
+        ],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+        cwd=REPO_ROOT,
+    )
+    session = tmp_path / "logs" / "interrupted"
+    # wait until telemetry is actually flowing (sqlite exists and grows)
+    deadline = _time.time() + 120
+    db = session / "aggregator" / "telemetry.sqlite"
+    while _time.time() < deadline and not db.exists():
+        _time.sleep(0.5)
+    assert db.exists(), proc.stderr and "no telemetry before interrupt"
+    _time.sleep(8)  # a few steps + at least one sampler tick
+
+    proc.send_signal(_signal.SIGINT)
+    out, err = proc.communicate(timeout=150)
+
+    summary_path = session / "final_summary.json"
+    assert summary_path.exists(), (out[-1500:], err[-2500:])
+    payload = json.loads(summary_path.read_text())
+    window = payload["step_time"]["global"]["window"]
+    assert (window["steps_analyzed"] or 0) > 0, "no steps in interrupted summary"
+    manifest = json.loads((session / "manifest.json").read_text())
+    assert manifest["status"] in ("completed", "failed")
