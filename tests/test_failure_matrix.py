@@ -509,3 +509,36 @@ def test_concurrent_clients_hammer_aggregator(tmp_path):
     finally:
         conn.close()
     assert not dupes, dupes[:5]
+
+
+@pytest.mark.timeout(60)
+def test_malformed_control_does_not_kill_ingest_loop(tmp_path):
+    """A control payload with junk meta previously raised inside the
+    aggregator's loop thread, silently stopping ALL telemetry. The bad
+    payload must be dropped and ingest must continue."""
+    agg = _aggregator(tmp_path, "hostile", finalize_timeout=6.0)
+    c = TCPClient("127.0.0.1", agg.port)
+    c.send_batch([
+        {"_traceml_control": "rank_finished", "meta": "junk-not-a-dict"},
+        {"_traceml_control": "rank_finished"},  # meta absent entirely
+    ])
+    time.sleep(0.3)
+    # the loop thread must still be alive and ingest must still work
+    assert agg._loop_thread.is_alive(), "ingest loop died on junk control"
+    for step in range(1, 6):
+        c.send_batch([_step_envelope(0, step)])
+    c.send_batch([build_rank_finished(_meta(0))])
+    deadline = time.time() + 15
+    count = 0
+    while time.time() < deadline and count < 5:
+        agg.sqlite.force_flush(2.0)
+        conn = sqlite3.connect(agg.db_path)
+        try:
+            count = conn.execute(
+                "SELECT COUNT(*) FROM step_time_samples"
+            ).fetchone()[0]
+        finally:
+            conn.close()
+        time.sleep(0.2)
+    agg.stop()
+    assert count == 5, "telemetry stopped flowing after junk control"

@@ -128,24 +128,32 @@ class TraceMLAggregator:
     def _drain_tcp(self) -> int:
         count = 0
         for payload in self.server.drain():
-            control = parse_control(payload)
-            if control is not None:
-                self._handle_control(control)
-                continue
-            envelope = normalize_telemetry_envelope(payload)
-            if envelope is None:
-                continue
-            meta = envelope["meta"]
-            rank = meta.get("global_rank")
-            if isinstance(rank, int):
-                self._seen_ranks.add(rank)
-            self.sqlite.ingest(envelope)
-            count += 1
+            # One hostile/corrupt payload must never kill the ingest loop
+            # thread (it would silently stop ALL telemetry): drop it.
+            try:
+                control = parse_control(payload)
+                if control is not None:
+                    self._handle_control(control)
+                    continue
+                envelope = normalize_telemetry_envelope(payload)
+                if envelope is None:
+                    continue
+                meta = envelope["meta"]
+                rank = meta.get("global_rank")
+                if isinstance(rank, int):
+                    self._seen_ranks.add(rank)
+                self.sqlite.ingest(envelope)
+                count += 1
+            except Exception:
+                logger.debug(
+                    "traceml_amd: malformed payload dropped", exc_info=True
+                )
         return count
 
     def _handle_control(self, control: dict) -> None:
         if control.get(CONTROL_KEY) == RANK_FINISHED:
-            rank = control.get("meta", {}).get("global_rank")
+            meta = control.get("meta")
+            rank = meta.get("global_rank") if isinstance(meta, dict) else None
             if isinstance(rank, int):
                 self._finished_ranks.add(rank)
 
