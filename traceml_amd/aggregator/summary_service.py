@@ -44,6 +44,8 @@ class FinalSummaryService:
                 request = json.load(f)
         except (OSError, ValueError):
             return False
+        if not isinstance(request, dict):
+            return False  # junk request file: ignore until overwritten
         request_id = request.get("request_id")
         if not request_id or request_id == self._last_request_id:
             return False
