@@ -66,7 +66,10 @@ class ProjectionWriter:
     def build_rows(self, envelope: dict) -> List[Tuple[str, dict]]:
         """envelope → [(sql_table, row_dict)]."""
         meta = envelope.get("meta", {})
-        identity = {k: meta.get(k) for k in IDENTITY_KEYS}
+        # identity values go through the same sanitizer as row values: a
+        # malformed meta (dict-valued rank) must not poison a whole flush
+        # batch at executemany time
+        identity = {k: _plain(meta.get(k)) for k in IDENTITY_KEYS}
         out: List[Tuple[str, dict]] = []
         for wire_table, rows in envelope.get("body", {}).get("tables", {}).items():
             spec = self.tables.get(wire_table)
