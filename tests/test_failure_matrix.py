@@ -542,3 +542,25 @@ def test_malformed_control_does_not_kill_ingest_loop(tmp_path):
         time.sleep(0.2)
     agg.stop()
     assert count == 5, "telemetry stopped flowing after junk control"
+
+
+def test_poisoned_row_does_not_drop_flush_batch(tmp_path):
+    """An unbindable value in ONE row (defense-in-depth beyond the
+    sanitizers) must not lose the other rows of the same flush."""
+    import sqlite3 as _sq
+
+    from traceml_amd.aggregator.writers import build_all_writers
+
+    writers = build_all_writers()
+    step_writer = next(w for w in writers if w.sampler == "step_time")
+    conn = _sq.connect(str(tmp_path / "p.sqlite"))
+    step_writer.init_schema(conn)
+    good = {"global_rank": 0, "timestamp": 1.0, "step": 1,
+            "events_json": "{}"}
+    poisoned = dict(good, step=object())  # unbindable python object
+    with conn:
+        step_writer.insert_rows(
+            conn, "step_time_samples", [good, poisoned, dict(good, step=2)]
+        )
+    n = conn.execute("SELECT COUNT(*) FROM step_time_samples").fetchone()[0]
+    assert n == 2  # the two good rows survived
