@@ -97,16 +97,21 @@ class ProjectionWriter:
         sql = (
             f"INSERT INTO {sql_table} ({', '.join(columns)}) VALUES ({placeholders})"
         )
+        conn.execute("SAVEPOINT traceml_ins")
         try:
             conn.executemany(sql, [[r.get(c) for c in columns] for r in rows])
         except sqlite3.Error:
-            # one unbindable row must not drop the whole flush batch:
-            # retry row-by-row and skip only the poisoned ones
+            # one unbindable row must not drop the whole flush batch: undo
+            # the partial executemany, then retry row-by-row and skip only
+            # the poisoned ones
+            conn.execute("ROLLBACK TO traceml_ins")
             for r in rows:
                 try:
                     conn.execute(sql, [r.get(c) for c in columns])
                 except sqlite3.Error:
                     continue
+        finally:
+            conn.execute("RELEASE traceml_ins")
 
     def sql_tables(self) -> List[str]:
         return [sql_table for sql_table, _ in self.tables.values()]
