@@ -279,13 +279,18 @@ def build_app(get_db_path, get_session=lambda: None):
     def index():
         return HTMLResponse(_PAGE)
 
+    # FastAPI runs sync endpoints in a threadpool: two concurrent /api/live
+    # polls would otherwise race on the shared LiveStepTimeSession cursor
+    poll_lock = threading.Lock()
+
     @app.get("/api/live")
     def live():
         db_path = get_db_path()
         if db_path is None:
             return JSONResponse({}, status_code=503)
         try:
-            return JSONResponse(_live_payload(db_path, get_session()))
+            with poll_lock:
+                return JSONResponse(_live_payload(db_path, get_session()))
         except Exception as exc:
             return JSONResponse({"error": repr(exc)}, status_code=500)
 
