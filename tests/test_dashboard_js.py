@@ -127,6 +127,26 @@ def test_dashboard_js_escapes_hostile_strings(tmp_path):
     assert "<img src=x" not in html
 
 
+def test_dashboard_js_model_health_banner(tmp_path):
+    """A memory-domain crit finding outranking the step-time verdict shows
+    the combined model-health banner."""
+    db_path = str(tmp_path / "t.sqlite")
+    scenarios.healthy_ddp(ranks=1, steps=30).write(db_path)
+    cap = 100 << 30
+    scenarios.write_memory_rows(
+        db_path, {0: (90 << 30, 98 << 30)}, capacity=cap, steps=10,
+    )
+    from traceml_amd.renderers import live_view
+
+    payload = live_view(db_path)
+    model = payload["sections"]["model"]["diagnosis"]
+    assert model["kind"] == "HIGH_MEMORY_PRESSURE"
+    assert model["evidence"]["domain"] == "step_memory"
+    html = _render(payload)
+    assert "[model health]" in html
+    assert "HIGH MEMORY PRESSURE" in html
+
+
 def test_dashboard_js_memory_sparkline(tmp_path):
     """Creeping memory draws a rising (amber) sparkline in the card."""
     db_path = str(tmp_path / "t.sqlite")

@@ -329,7 +329,7 @@ def test_live_view_sections_complete(db_path):
     payload = live_view(db_path)
     sections = payload["sections"]
     assert set(sections) == {
-        "step_time", "step_memory", "system", "process", "comm",
+        "step_time", "step_memory", "system", "process", "comm", "model",
     }
     for name, view in sections.items():
         assert view["section"] == name
@@ -433,7 +433,7 @@ def test_dashboard_api_contract(db_path):
     assert live.status_code == 200
     payload = live.json()
     assert set(payload["sections"]) == {
-        "step_time", "step_memory", "system", "process", "comm",
+        "step_time", "step_memory", "system", "process", "comm", "model",
     }
     st = payload["sections"]["step_time"]
     assert st["diagnosis"]["kind"] == "INPUT_STRAGGLER"

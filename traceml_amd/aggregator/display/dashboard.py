@@ -246,6 +246,14 @@ async function tick(){
     const s=d.sections||{};
     let html='';
     html+=stepTimeSection(s.step_time);
+    // model-health banner: surface a memory-domain finding that outranks
+    // the step-time verdict (combined model-diagnostics card)
+    if(s.model&&s.model.diagnosis&&s.model.diagnosis.evidence&&
+       s.model.diagnosis.evidence.domain==='step_memory'&&
+       (s.model.diagnosis.severity==='warn'||s.model.diagnosis.severity==='crit')){
+      const m=s.model.diagnosis;
+      html+=`<div class="verdict ${m.severity}" style="font-weight:500">${esc(m.status)} <span class="dim">[model health]</span> ${esc(m.summary)}</div>`;
+    }
     html+=memorySection(s.step_memory);
     html+=commSection(s.comm);
     html+=findings(d);

@@ -173,17 +173,32 @@ def live_view(db_path: str, session=None) -> dict:
     from traceml_amd.renderers.step_time import render_step_time
     from traceml_amd.renderers.system import render_system
 
+    from traceml_amd.diagnostics.model_diagnostics import (
+        compose_model_diagnostics,
+    )
+
     memory_series = load_memory_series(db_path)
+    memory_diagnosis = diagnose_step_memory(memory_series)
     system_ctx = load_system_context(db_path)
     process_ctx = load_process_context(db_path)
+    model_combined = compose_model_diagnostics(
+        result.diagnosis, memory_diagnosis
+    )
     sections = {
         "step_time": render_step_time(result.window, result.diagnosis),
-        "step_memory": render_step_memory(
-            memory_series, diagnose_step_memory(memory_series)
-        ),
+        "step_memory": render_step_memory(memory_series, memory_diagnosis),
         "system": render_system(system_ctx, diagnose_system(system_ctx)),
         "process": render_process(process_ctx, diagnose_process(process_ctx)),
         "comm": render_comm(load_latest_gather(db_path)),
+        # combined model-health card: most severe of step_time/step_memory
+        # with the originating domain in evidence (the reference's
+        # model-diagnostics dashboard card)
+        "model": {
+            "section": "model",
+            "available": result.window.has_data,
+            "diagnosis": model_combined.primary.to_payload(),
+            "issues": [i.to_payload() for i in model_combined.issues],
+        },
     }
     payload = {
         "freshness": freshness,
