@@ -235,3 +235,21 @@ def test_exporter_final_drain_budget(monkeypatch):
     elapsed = _time.time() - start
     assert elapsed < 10.0  # bounded, not 50*0.05 + unbounded wait
     assert client.sent  # at least part of the backlog flushed
+
+
+def test_compat_alias_delegates_with_deprecation():
+    """`import traceml` (reference package name) warns once and delegates
+    every public symbol to traceml_amd (PARITY row 69)."""
+    import importlib
+    import warnings
+
+    import traceml_amd
+
+    with warnings.catch_warnings(record=True) as caught:
+        warnings.simplefilter("always")
+        import traceml
+        importlib.reload(traceml)
+    assert any(issubclass(w.category, DeprecationWarning) for w in caught)
+    assert traceml.init is traceml_amd.init
+    assert traceml.trace_step is traceml_amd.trace_step
+    assert traceml.summary is traceml_amd.summary
