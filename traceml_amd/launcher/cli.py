@@ -64,6 +64,11 @@ def build_parser() -> argparse.ArgumentParser:
         prog="traceml-amd",
         description="MI355X-native training-step profiler",
     )
+    from traceml_amd.version import __version__
+
+    parser.add_argument(
+        "--version", action="version", version=f"traceml-amd {__version__}"
+    )
     sub = parser.add_subparsers(dest="command", required=True)
 
     run_p = sub.add_parser("run", help="profile a training script (summary mode)")
