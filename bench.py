@@ -153,13 +153,6 @@ class Workload:
             num_workers=0,
         )
 
-    def config_json(self):
-        return {
-            "model": self.args.model,
-            "global_batch": self.batch,
-            "seq_len": self.seq,
-        }
-
 
 class ResNetWorkload(Workload):
     """BASELINE config 2: ResNet-50 bf16, bs 256/GPU, synthetic ImageNet."""
