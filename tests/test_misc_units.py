@@ -253,3 +253,35 @@ def test_compat_alias_delegates_with_deprecation():
     assert traceml.init is traceml_amd.init
     assert traceml.trace_step is traceml_amd.trace_step
     assert traceml.summary is traceml_amd.summary
+
+
+def test_sampler_registry_rank_and_mode_gating():
+    """Declarative sampler policy: system + stdout are local-rank-0 only,
+    stdout only in live display modes, everything else on every rank
+    (reference: runtime/sampler_registry.py:78-125)."""
+    from traceml_amd.runtime.identity import RuntimeIdentity
+    from traceml_amd.runtime.registry import build_samplers
+
+    def names(local_rank, mode):
+        identity = RuntimeIdentity(
+            global_rank=local_rank, local_rank=local_rank, world_size=8,
+            local_world_size=8, node_rank=0, hostname="h", pid=1,
+        )
+        return {spec.name for spec, _s, _db in build_samplers(identity, mode)}
+
+    rank0_cli = names(0, "cli")
+    rank3_cli = names(3, "cli")
+    rank0_summary = names(0, "summary")
+
+    assert "system" in rank0_cli and "system" not in rank3_cli
+    assert "stdout_stderr" in rank0_cli and "stdout_stderr" not in rank3_cli
+    assert "stdout_stderr" not in rank0_summary  # live display modes only
+    for always in ("step_time", "step_memory", "process", "rank_stats",
+                   "runtime_environment"):
+        assert always in rank0_cli and always in rank3_cli
+    # each sampler owns an independent database (cursor isolation)
+    identity = RuntimeIdentity(global_rank=0, local_rank=0, world_size=1,
+                               local_world_size=1, node_rank=0, hostname="h",
+                               pid=1)
+    dbs = [db for _s, _x, db in build_samplers(identity, "cli")]
+    assert len({id(db) for db in dbs}) == len(dbs)
