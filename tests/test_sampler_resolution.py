@@ -155,3 +155,49 @@ def test_step_time_queue_drop_oldest(monkeypatch):
     batches = timing.drain_step_time_queue()
     assert [b.step for b in batches] == [6, 7, 8, 9, 10]
     timing.clear_for_tests()
+
+
+def test_aggregate_batch_semantics():
+    """Per-step aggregation: duration uses the GPU clock EXCEPT for the
+    CPU-clock-preferred events (dataloader, step envelope); multi-call
+    events sum; a lost GPU side falls back to CPU for that event."""
+    from traceml_amd.core import event_names, timing
+    from traceml_amd.samplers.step_time import aggregate_batch
+
+    def ev(name, cpu_ms, gpu_ms=None):
+        e = timing.TimeEvent(name=name, device="cuda" if gpu_ms else "cpu",
+                             cpu_start=0.0, cpu_end=cpu_ms / 1000.0)
+        e.gpu_ms = gpu_ms
+        e._gpu_done = True
+        return e
+
+    batch = timing.StepTimeBatch(
+        step=7,
+        flushed_at=123.0,
+        events=[
+            ev(event_names.DATALOADER, 50.0, gpu_ms=None),
+            ev(event_names.FORWARD, 10.0, gpu_ms=12.0),
+            ev(event_names.FORWARD, 11.0, gpu_ms=13.0),   # 2 calls sum
+            ev(event_names.BACKWARD, 20.0, gpu_ms=None),  # lost GPU side
+            ev(event_names.STEP_TIME, 100.0, gpu_ms=140.0),
+        ],
+    )
+    row = aggregate_batch(batch)
+    assert row["step"] == 7 and row["timestamp"] == 123.0
+    events = row["events"]
+    # dataloader: CPU-clock preferred even though is_gpu False anyway
+    assert events[event_names.DATALOADER]["duration_ms"] == pytest.approx(50.0)
+    # forward: GPU clock, summed across calls, n_calls=2
+    fwd = events[event_names.FORWARD]
+    assert fwd["duration_ms"] == pytest.approx(25.0)
+    assert fwd["gpu_ms"] == pytest.approx(25.0)
+    assert fwd["cpu_ms"] == pytest.approx(21.0)
+    assert fwd["n_calls"] == 2 and fwd["is_gpu"] is True
+    # backward lost its GPU pair -> CPU fallback for duration, is_gpu False
+    bwd = events[event_names.BACKWARD]
+    assert bwd["duration_ms"] == pytest.approx(20.0)
+    assert bwd["gpu_ms"] is None and bwd["is_gpu"] is False
+    # step envelope: CPU-clock preferred (the wall bracket), GPU kept as data
+    st = events[event_names.STEP_TIME]
+    assert st["duration_ms"] == pytest.approx(100.0)
+    assert st["gpu_ms"] == pytest.approx(140.0)
